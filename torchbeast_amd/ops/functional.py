@@ -1,0 +1,270 @@
+"""Autograd wrappers for the fused CDNA4 kernels, with eager oracles.
+
+Every op has two implementations:
+- the HIP one (extension `_tbops`, gfx950 kernels), used for CUDA tensors;
+- an eager PyTorch one, used on CPU and as the fp32 numerics oracle in tests.
+
+On a GPU machine the eager path is NOT silently substituted: if the
+extension lacks a kernel the op raises (TBAMD_ALLOW_EAGER=1 overrides).
+"""
+
+import os
+
+import torch
+import torch.nn.functional as F
+
+import torchbeast_amd.ops as ops_mod
+
+
+def _ext_for(tensor, symbol):
+    """Return the extension if `tensor` is on GPU and `symbol` exists."""
+    if not tensor.is_cuda:
+        return None
+    ext = ops_mod.require_ext() if not os.environ.get("TBAMD_ALLOW_EAGER") else ops_mod._load()
+    if ext is not None and hasattr(ext, symbol):
+        return ext
+    if os.environ.get("TBAMD_ALLOW_EAGER"):
+        return None
+    raise RuntimeError(
+        f"HIP kernel '{symbol}' missing from _tbops on a GPU tensor "
+        "(rebuild the extension, or set TBAMD_ALLOW_EAGER=1)"
+    )
+
+
+# ---------------------------------------------------------------------------
+# V-trace (fused): action log-probs + rho/c clip + reverse scan + advantages.
+# ---------------------------------------------------------------------------
+
+
+def vtrace_from_logits(
+    behavior_policy_logits,
+    target_policy_logits,
+    actions,
+    discounts,
+    rewards,
+    values,
+    bootstrap_value,
+    clip_rho_threshold=1.0,
+    clip_pg_rho_threshold=1.0,
+):
+    from torchbeast_amd.core import vtrace as pyvtrace
+
+    ext = _ext_for(target_policy_logits, "vtrace_from_logits")
+    if ext is None:
+        # Eager path (also the oracle): delegate to the pure-PyTorch module.
+        target_lp = pyvtrace.action_log_probs(target_policy_logits, actions)
+        behavior_lp = pyvtrace.action_log_probs(behavior_policy_logits, actions)
+        log_rhos = target_lp - behavior_lp
+        core = pyvtrace.from_importance_weights(
+            log_rhos,
+            discounts,
+            rewards,
+            values,
+            bootstrap_value,
+            clip_rho_threshold,
+            clip_pg_rho_threshold,
+        )
+        return pyvtrace.VTraceFromLogitsReturns(
+            vs=core.vs,
+            pg_advantages=core.pg_advantages,
+            log_rhos=log_rhos,
+            behavior_action_log_probs=behavior_lp,
+            target_action_log_probs=target_lp,
+        )
+
+    with torch.no_grad():
+        vs, pg_adv, log_rhos, blp, tlp = ext.vtrace_from_logits(
+            behavior_policy_logits.detach().float().contiguous(),
+            target_policy_logits.detach().float().contiguous(),
+            actions.contiguous(),
+            discounts.detach().float().contiguous(),
+            rewards.detach().float().contiguous(),
+            values.detach().float().contiguous(),
+            bootstrap_value.detach().float().contiguous(),
+            float(clip_rho_threshold),
+            float(clip_pg_rho_threshold),
+        )
+    return pyvtrace.VTraceFromLogitsReturns(
+        vs=vs,
+        pg_advantages=pg_adv,
+        log_rhos=log_rhos,
+        behavior_action_log_probs=blp,
+        target_action_log_probs=tlp,
+    )
+
+
+# ---------------------------------------------------------------------------
+# Fused IMPALA loss: pg + baseline + entropy, with analytic gradients.
+# ---------------------------------------------------------------------------
+
+
+class _FusedImpalaLoss(torch.autograd.Function):
+    """Forward computes the three scalar losses AND the gradients w.r.t.
+    (logits, baseline) in one kernel pass; backward just scales the saved
+    gradients by the incoming scalar grads. grads:
+      d pg/d logits       = (softmax - onehot(a)) * pg_adv        (adv detached)
+      d entropy/d logits  = softmax * (logp - sum(p*logp))
+      d baseline/d bl     = baseline - vs
+    """
+
+    @staticmethod
+    def forward(ctx, logits, baseline, actions, pg_advantages, vs):
+        ext = _ext_for(logits, "fused_impala_loss_fwd")
+        (
+            pg_loss,
+            baseline_loss,
+            entropy_loss,
+            d_logits_pg,
+            d_logits_ent,
+            d_baseline,
+        ) = ext.fused_impala_loss_fwd(
+            logits.detach().float().contiguous(),
+            baseline.detach().float().contiguous(),
+            actions.contiguous(),
+            pg_advantages.detach().float().contiguous(),
+            vs.detach().float().contiguous(),
+        )
+        ctx.save_for_backward(d_logits_pg, d_logits_ent, d_baseline)
+        return pg_loss, baseline_loss, entropy_loss
+
+    @staticmethod
+    def backward(ctx, g_pg, g_bl, g_ent):
+        d_logits_pg, d_logits_ent, d_baseline = ctx.saved_tensors
+        grad_logits = g_pg * d_logits_pg + g_ent * d_logits_ent
+        grad_baseline = g_bl * d_baseline
+        return grad_logits, grad_baseline, None, None, None
+
+
+def fused_impala_loss(logits, baseline, actions, pg_advantages, vs):
+    """Return (pg_loss, baseline_loss, entropy_loss) scalars.
+
+    baseline_loss here is 0.5*sum((vs - baseline)^2) — the caller applies the
+    cost weights.
+    """
+    if logits.is_cuda and _ext_for(logits, "fused_impala_loss_fwd") is not None:
+        return _FusedImpalaLoss.apply(logits, baseline, actions, pg_advantages, vs)
+
+    from torchbeast_amd.core import losses
+
+    pg_loss = losses.compute_policy_gradient_loss(logits, actions, pg_advantages)
+    baseline_loss = losses.compute_baseline_loss(vs - baseline)
+    entropy_loss = losses.compute_entropy_loss(logits)
+    return pg_loss, baseline_loss, entropy_loss
+
+
+# ---------------------------------------------------------------------------
+# Done-masked LSTM unroll over T (multi-layer).
+# ---------------------------------------------------------------------------
+
+
+class _LstmUnroll(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, notdone, h0, c0, *flat_weights):
+        ext = _ext_for(x, "lstm_unroll_fwd")
+        num_layers = len(flat_weights) // 4
+        out, hT, cT, stash = ext.lstm_unroll_fwd(
+            x.contiguous(),
+            notdone.contiguous(),
+            h0.contiguous(),
+            c0.contiguous(),
+            list(flat_weights),
+        )
+        ctx.save_for_backward(x, notdone, h0, c0, stash, *flat_weights)
+        ctx.num_layers = num_layers
+        return out, hT, cT
+
+    @staticmethod
+    def backward(ctx, d_out, d_hT, d_cT):
+        x, notdone, h0, c0, stash, *flat_weights = ctx.saved_tensors
+        ext = ops_mod.require_ext()
+        grads = ext.lstm_unroll_bwd(
+            x,
+            notdone,
+            h0,
+            c0,
+            stash,
+            list(flat_weights),
+            d_out.contiguous(),
+            d_hT.contiguous(),
+            d_cT.contiguous(),
+        )
+        d_x, d_h0, d_c0, *d_weights = grads
+        return (d_x, None, d_h0, d_c0, *d_weights)
+
+
+def lstm_unroll(core, x, notdone, state):
+    """Run `core` (an nn.LSTM) over x [T,B,I] with per-step done masking.
+
+    state is (h, c), each [L, B, H]. Mirrors the model-level Python loop:
+        state <- notdone_t * state;  out_t, state <- lstm(x_t, state)
+    Returns (output [T,B,H], (hT, cT)).
+    """
+    if x.is_cuda and _ext_for(x, "lstm_unroll_fwd") is not None:
+        flat = []
+        for layer in range(core.num_layers):
+            flat += [
+                getattr(core, f"weight_ih_l{layer}"),
+                getattr(core, f"weight_hh_l{layer}"),
+                getattr(core, f"bias_ih_l{layer}"),
+                getattr(core, f"bias_hh_l{layer}"),
+            ]
+        out, hT, cT = _LstmUnroll.apply(x, notdone, state[0], state[1], *flat)
+        return out, (hT, cT)
+
+    outputs = []
+    for xt, nd in zip(x.unbind(), notdone.unbind()):
+        nd = nd.view(1, -1, 1)
+        state = tuple(nd * s for s in state)
+        out, state = core(xt.unsqueeze(0), state)
+        outputs.append(out)
+    return torch.cat(outputs), state
+
+
+# ---------------------------------------------------------------------------
+# Fused grad-clip + RMSProp (+ optional LR from a lambda schedule).
+# ---------------------------------------------------------------------------
+
+
+def rmsprop_step(param, grad, square_avg, lr, alpha, eps, clip_norm=None):
+    """One RMSProp step over FLAT tensors, fused with global-norm clipping.
+
+    All of param/grad/square_avg are 1-D views over the whole model (see
+    parallel.flat_params). Returns the pre-clip gradient norm (0-dim tensor).
+    """
+    ext = _ext_for(param, "rmsprop_step") if param.is_cuda else None
+    if ext is not None:
+        return ext.rmsprop_step(
+            param, grad, square_avg, float(lr), float(alpha), float(eps),
+            float(clip_norm) if clip_norm is not None else -1.0,
+        )
+
+    total_norm = grad.norm(2)
+    if clip_norm is not None:
+        coef = clip_norm / (total_norm + 1e-6)
+        if coef < 1:
+            grad = grad.mul(coef)
+    square_avg.mul_(alpha).addcmul_(grad, grad, value=1 - alpha)
+    avg = square_avg.sqrt().add_(eps)
+    param.addcdiv_(grad, avg, value=-lr)
+    return total_norm
+
+
+# ---------------------------------------------------------------------------
+# Policy sampling: softmax + multinomial (train) / argmax (eval).
+# ---------------------------------------------------------------------------
+
+
+def policy_sample(policy_logits, greedy=False, generator=None):
+    """Sample actions from categorical logits [N, A] -> [N] int64."""
+    if policy_logits.is_cuda:
+        ext = _ext_for(policy_logits, "policy_sample")
+        if ext is not None:
+            seed = int(torch.randint(0, 2**62, (1,)).item()) if not greedy else 0
+            return ext.policy_sample(
+                policy_logits.detach().float().contiguous(), bool(greedy), seed
+            )
+    if greedy:
+        return torch.argmax(policy_logits, dim=-1)
+    return torch.multinomial(
+        F.softmax(policy_logits, dim=-1), num_samples=1, generator=generator
+    ).squeeze(-1)
