@@ -1,0 +1,58 @@
+"""Build the native extensions in-tree (gfx950 / MI355X).
+
+- torchbeast_amd.runtime._tbruntime : C++ actor-learner runtime (queues,
+  dynamic batcher, actor pool, unix-socket env server, nest).
+- torchbeast_amd.ops._tbops         : hand-written CDNA4 HIP kernels
+  (V-trace, fused losses, LSTM unroll, RMSProp, sampling, conv trunk).
+
+Usage: PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+"""
+
+import glob
+import os
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CppExtension, CUDAExtension
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+
+ext_modules = [
+    CppExtension(
+        name="torchbeast_amd.runtime._tbruntime",
+        sources=["torchbeast_amd/runtime/csrc/module.cc"],
+        extra_compile_args=["-O3", "-std=c++17", "-pthread"],
+    )
+]
+
+hip_sources = sorted(glob.glob("torchbeast_amd/ops/hip/*.hip"))
+if hip_sources:
+    ext_modules.append(
+        CUDAExtension(
+            name="torchbeast_amd.ops._tbops",
+            sources=hip_sources,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    )
+
+setup(
+    name="torchbeast_amd",
+    version="0.1.0",
+    description="MI355X-native IMPALA actor-learner framework",
+    packages=[
+        "torchbeast_amd",
+        "torchbeast_amd.core",
+        "torchbeast_amd.models",
+        "torchbeast_amd.envs",
+        "torchbeast_amd.ops",
+        "torchbeast_amd.runtime",
+        "torchbeast_amd.nest",
+        "torchbeast_amd.parallel",
+    ],
+    ext_modules=ext_modules,
+    cmdclass={"build_ext": BuildExtension},
+)

@@ -19,11 +19,16 @@ from torch.nn import functional as F
 
 
 class AtariNet(nn.Module):
-    def __init__(self, observation_shape, num_actions, use_lstm=False):
+    def __init__(self, observation_shape, num_actions, use_lstm=False,
+                 use_last_action=True):
         super().__init__()
         self.observation_shape = observation_shape
         self.num_actions = num_actions
         self.use_lstm = use_lstm
+        # The polybeast env plane doesn't transport last_action (the deep
+        # reference net doesn't consume it either, ref:
+        # polybeast_learner.py:234-235); monobeast's AtariNet does.
+        self.use_last_action = use_last_action
 
         in_channels = observation_shape[0]
         self.conv1 = nn.Conv2d(in_channels, 32, kernel_size=8, stride=4)
@@ -34,8 +39,8 @@ class AtariNet(nn.Module):
         conv_out = self._conv_out_elems(observation_shape)
         self.fc = nn.Linear(conv_out, 512)
 
-        # FC output ⊕ one-hot last action ⊕ clipped reward.
-        core_size = self.fc.out_features + num_actions + 1
+        # FC output ⊕ clipped reward ⊕ (optional) one-hot last action.
+        core_size = self.fc.out_features + 1 + (num_actions if use_last_action else 0)
         if use_lstm:
             self.core = nn.LSTM(core_size, core_size, num_layers=2)
 
@@ -65,10 +70,13 @@ class AtariNet(nn.Module):
         x = F.relu(self.conv3(x))
         x = F.relu(self.fc(x.view(T * B, -1)))
 
-        last_action = torch.flatten(inputs["last_action"], 0, 1)
-        one_hot_action = F.one_hot(last_action, self.num_actions).float()
         clipped_reward = torch.clamp(inputs["reward"], -1, 1).view(T * B, 1)
-        core_input = torch.cat([x, clipped_reward, one_hot_action], dim=-1)
+        if self.use_last_action:
+            last_action = torch.flatten(inputs["last_action"], 0, 1)
+            one_hot_action = F.one_hot(last_action, self.num_actions).float()
+            core_input = torch.cat([x, clipped_reward, one_hot_action], dim=-1)
+        else:
+            core_input = torch.cat([x, clipped_reward], dim=-1)
 
         if self.use_lstm:
             core_input = core_input.view(T, B, -1)

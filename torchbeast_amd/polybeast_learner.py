@@ -1,0 +1,470 @@
+"""PolyBeast learner (ref: torchbeast/polybeast_learner.py) — MI355X-native.
+
+One process per GPU (launched directly for 1 GPU, or via
+`python -m torch.distributed.run --nproc-per-node N` for data-parallel
+training over RCCL/xGMI). Each rank owns:
+
+- a C++ ActorPool driving `--num_actors` environment streams (unix-socket
+  env servers, or in-process native synthetic envs for `--env synthetic*`),
+- a DynamicBatcher feeding batched behavior-model inference on the GPU,
+- a BatchingQueue of [T+1, B] rollouts feeding the learner step
+  (V-trace + IMPALA losses + fused clip/RMSProp on a flat parameter buffer),
+- with world_size > 1, a single flat-gradient RCCL all-reduce per step.
+
+Differences from the reference runtime (same capabilities, MI355X design):
+queues hand over pinned host tensors (DMA H2D), the optimizer/clip/LR are
+one fused kernel over a flat buffer, and weight sync to the behavior model
+is one flat device copy instead of a state_dict walk.
+"""
+
+import argparse
+import collections
+import logging
+import os
+import threading
+import time
+import timeit
+
+import torch
+
+from torchbeast_amd import flags as tbflags
+from torchbeast_amd import runtime
+from torchbeast_amd.core import file_writer, vtrace
+from torchbeast_amd.models.atari_net import AtariNet
+from torchbeast_amd.models.resnet import ResNet
+from torchbeast_amd.ops import functional as tbops
+from torchbeast_amd.parallel import ddp as tbddp
+from torchbeast_amd.parallel import flat as tbflat
+
+logging.basicConfig(
+    format="[%(levelname)s:%(process)d %(module)s:%(lineno)d %(asctime)s] %(message)s",
+    level=0,
+)
+
+EnvOutput = collections.namedtuple(
+    "EnvOutput", "frame rewards done episode_step episode_return"
+)
+AgentOutput = collections.namedtuple("AgentOutput", "action policy_logits baseline")
+
+
+def make_parser():
+    parser = argparse.ArgumentParser(description="MI355X-native PolyBeast learner")
+    parser.add_argument("--mode", default="train", choices=["train", "test"])
+    tbflags.add_common_flags(parser)
+    parser.add_argument("--pipes_basename", default="unix:/tmp/polybeast",
+                        help="Basename for the env-server unix sockets.")
+    parser.add_argument("--num_actors", default=4, type=int,
+                        help="Env streams driven by this rank's ActorPool.")
+    parser.add_argument("--num_learner_threads", default=2, type=int)
+    parser.add_argument("--num_inference_threads", default=2, type=int)
+    parser.add_argument("--max_learner_queue_size", default=None, type=int)
+    parser.add_argument("--num_actions", default=6, type=int)
+    parser.add_argument("--model", default="shallow", choices=["shallow", "deep"],
+                        help="shallow=AtariNet (the headline bench model), "
+                             "deep=IMPALA ResNet.")
+    parser.add_argument("--inference_max_batch_size", default=512, type=int)
+    parser.add_argument("--inference_timeout_ms", default=10, type=int)
+    parser.add_argument("--write_profiler_trace", action="store_true")
+    tbflags.add_loss_flags(parser)
+    tbflags.add_optimizer_flags(parser)
+    return parser
+
+
+parser = make_parser()
+
+
+def observation_shape(flags):
+    spec = tbflags.parse_synthetic_env_spec(flags.env)
+    if spec is not None:
+        return spec[0], spec[1]
+    return (4, 84, 84), flags.num_actions
+
+
+def create_model(flags):
+    shape, num_actions = observation_shape(flags)
+    if flags.model == "shallow":
+        return AtariNet(shape, num_actions, use_lstm=flags.use_lstm,
+                        use_last_action=False)
+    return ResNet(shape, num_actions, use_lstm=flags.use_lstm)
+
+
+def _as_agent_output(outputs):
+    """Normalize model output (AtariNet dict / ResNet tuple) to a tuple."""
+    if isinstance(outputs, dict):
+        return (outputs["action"], outputs["policy_logits"], outputs["baseline"])
+    return tuple(outputs)
+
+
+def inference(flags, inference_batcher, model, lock=threading.Lock()):  # noqa: B008
+    """Consume inference batches: one batched behavior-model forward each."""
+    with torch.no_grad():
+        for batch in inference_batcher:
+            batched_env_outputs, agent_state = batch.get_inputs()
+            frame, reward, done, *_ = batched_env_outputs
+            frame = frame.to(flags.actor_device, non_blocking=True)
+            reward = reward.to(flags.actor_device, non_blocking=True)
+            done = done.to(flags.actor_device, non_blocking=True)
+            agent_state = tuple(
+                t.to(flags.actor_device, non_blocking=True) for t in agent_state
+            )
+            with lock:
+                outputs, new_state = model(
+                    dict(frame=frame, reward=reward, done=done), agent_state
+                )
+            outputs = _as_agent_output(outputs)
+            outputs = tuple(t.cpu() for t in outputs)
+            new_state = tuple(t.cpu() for t in new_state)
+            batch.set_outputs((outputs, new_state))
+
+
+def learn(
+    flags,
+    learner_queue,
+    model,
+    flat_param,
+    flat_grad,
+    actor_flat,
+    optimizer,
+    scheduler,
+    stats,
+    plogger,
+    reducer,
+    num_updates,
+    update_counter,
+    lock=threading.Lock(),  # noqa: B008
+):
+    """Learner loop: exactly num_updates optimizer steps (shared across
+    threads via update_counter), identical on every DP rank so collectives
+    stay aligned."""
+    for tensors in learner_queue:
+        tensors = tuple(
+            t.to(flags.learner_device, non_blocking=True)
+            for t in runtime._tbruntime.flatten(tensors)
+        )
+        # Structure: ((env_outputs[5], actor_outputs[3]), initial_agent_state).
+        env_outputs = EnvOutput._make(tensors[:5])
+        actor_outputs = AgentOutput._make(tensors[5:8])
+        initial_agent_state = tensors[8:]
+
+        with lock:
+            with update_counter["mutex"]:
+                if update_counter["done"] >= num_updates:
+                    return
+                update_counter["done"] += 1
+
+            learner_outputs, _ = model(
+                dict(
+                    frame=env_outputs.frame,
+                    reward=env_outputs.rewards,
+                    done=env_outputs.done,
+                ),
+                initial_agent_state,
+            )
+            learner_outputs = AgentOutput._make(_as_agent_output(learner_outputs))
+
+            bootstrap_value = learner_outputs.baseline[-1]
+
+            # Shift: env_outputs[t+1] is the consequence of actions[t].
+            env_outputs = EnvOutput._make(t[1:] for t in env_outputs)
+            actor_outputs = AgentOutput._make(t[1:] for t in actor_outputs)
+            learner_outputs = AgentOutput._make(t[:-1] for t in learner_outputs)
+
+            if flags.reward_clipping == "abs_one":
+                clipped_rewards = torch.clamp(env_outputs.rewards, -1, 1)
+            else:
+                clipped_rewards = env_outputs.rewards
+
+            discounts = (~env_outputs.done).float() * flags.discounting
+
+            vtrace_returns = vtrace.from_logits(
+                behavior_policy_logits=actor_outputs.policy_logits,
+                target_policy_logits=learner_outputs.policy_logits,
+                actions=actor_outputs.action,
+                discounts=discounts,
+                rewards=clipped_rewards,
+                values=learner_outputs.baseline,
+                bootstrap_value=bootstrap_value,
+            )
+
+            pg_loss, baseline_loss, entropy_loss = tbops.fused_impala_loss(
+                learner_outputs.policy_logits,
+                learner_outputs.baseline,
+                actor_outputs.action,
+                vtrace_returns.pg_advantages,
+                vtrace_returns.vs,
+            )
+            total_loss = (
+                pg_loss
+                + flags.baseline_cost * baseline_loss
+                + flags.entropy_cost * entropy_loss
+            )
+
+            optimizer.zero_grad()
+            total_loss.backward()
+            reducer.reduce()
+            optimizer.step()
+            scheduler.step()
+
+            # Behavior-model sync: one flat copy (D2D on GPU).
+            with torch.no_grad():
+                actor_flat.copy_(flat_param)
+
+            episode_returns = env_outputs.episode_return[env_outputs.done]
+            stats["step"] = stats.get("step", 0) + flags.unroll_length * flags.batch_size
+            stats["episode_returns"] = tuple(episode_returns.cpu().numpy())
+            stats["mean_episode_return"] = torch.mean(episode_returns).item()
+            stats["mean_episode_step"] = torch.mean(
+                env_outputs.episode_step.float()
+            ).item()
+            stats["total_loss"] = total_loss.item()
+            stats["pg_loss"] = pg_loss.item()
+            stats["baseline_loss"] = baseline_loss.item()
+            stats["entropy_loss"] = entropy_loss.item()
+            stats["learner_queue_size"] = learner_queue.size()
+
+            if plogger is not None:
+                plogger.log(stats)
+
+            if not len(episode_returns):
+                stats["mean_episode_return"] = None
+
+
+def train(flags):  # noqa: C901
+    rank, world_size, local_rank = tbddp.maybe_init_distributed()
+    is_leader = rank == 0
+
+    if flags.xpid is None:
+        flags.xpid = "polybeast-%s" % time.strftime("%Y%m%d-%H%M%S")
+    plogger = None
+    if is_leader:
+        plogger = file_writer.FileWriter(
+            xpid=flags.xpid, xp_args=flags.__dict__, rootdir=flags.savedir
+        )
+    checkpointpath = os.path.expandvars(
+        os.path.expanduser("%s/%s/%s" % (flags.savedir, flags.xpid, "model.tar"))
+    )
+
+    if not flags.disable_cuda and torch.cuda.is_available():
+        flags.learner_device = torch.device("cuda", local_rank)
+        flags.actor_device = torch.device("cuda", local_rank)
+        torch.cuda.set_device(local_rank)
+    else:
+        flags.learner_device = torch.device("cpu")
+        flags.actor_device = torch.device("cpu")
+
+    if flags.max_learner_queue_size is None:
+        flags.max_learner_queue_size = flags.batch_size
+
+    # Rollouts are batched along dim 1 ([T+1, B, ...]).
+    learner_queue = runtime.BatchingQueue(
+        batch_dim=1,
+        minimum_batch_size=flags.batch_size,
+        maximum_batch_size=flags.batch_size,
+        check_inputs=True,
+        maximum_queue_size=flags.max_learner_queue_size,
+    )
+    inference_batcher = runtime.DynamicBatcher(
+        batch_dim=1,
+        minimum_batch_size=1,
+        maximum_batch_size=flags.inference_max_batch_size,
+        timeout_ms=flags.inference_timeout_ms,
+        check_outputs=True,
+    )
+
+    if tbflags.parse_synthetic_env_spec(flags.env) is not None:
+        # In-process native synthetic envs; give each rank distinct streams.
+        addresses = [
+            f"{flags.env}" for _ in range(flags.num_actors)
+        ]
+    else:
+        addresses = []
+        basename = flags.pipes_basename
+        for i in range(flags.num_actors):
+            addresses.append(f"{basename}.{rank * flags.num_actors + i}")
+
+    model = create_model(flags).to(device=flags.learner_device)
+    actor_model = create_model(flags).to(device=flags.actor_device)
+
+    flat_param = tbflat.flatten_parameters(model)
+    flat_grad = tbflat.attach_flat_grads(model)
+    actor_flat = tbflat.flatten_parameters(actor_model)
+
+    # All ranks start from rank 0's init.
+    tbddp.broadcast_flat(flat_param)
+    with torch.no_grad():
+        actor_flat.copy_(flat_param)
+
+    optimizer = tbflat.FusedRMSProp(
+        flat_param,
+        flat_grad,
+        lr=flags.learning_rate,
+        alpha=flags.alpha,
+        eps=flags.epsilon,
+        clip_norm=flags.grad_norm_clipping,
+    )
+    steps_per_update = flags.unroll_length * flags.batch_size * world_size
+    scheduler = tbflat.LinearLR(optimizer, steps_per_update, flags.total_steps)
+    reducer = tbddp.GradAllReducer(flat_grad, world_size)
+
+    stats = {}
+
+    if getattr(flags, "checkpoint_exists_ok", True) and os.path.exists(
+        checkpointpath
+    ):
+        logging.info("Resuming from checkpoint %s", checkpointpath)
+        ckpt = torch.load(checkpointpath, map_location=flags.learner_device,
+                          weights_only=False)
+        model.load_state_dict(ckpt["model_state_dict"])
+        optimizer.load_state_dict(ckpt["optimizer_state_dict"])
+        scheduler.load_state_dict(ckpt["scheduler_state_dict"])
+        stats = ckpt.get("stats", {})
+        with torch.no_grad():
+            actor_flat.copy_(flat_param)
+
+    initial_agent_state = model.initial_state(batch_size=1)
+    initial_agent_state = tuple(t.cpu() for t in initial_agent_state)
+
+    actor_pool = runtime.ActorPool(
+        unroll_length=flags.unroll_length,
+        learner_queue=learner_queue,
+        inference_batcher=inference_batcher,
+        env_server_addresses=addresses,
+        initial_agent_state=initial_agent_state,
+    )
+
+    def run_pool():
+        try:
+            actor_pool.run()
+        except Exception:
+            logging.exception("Exception in actor pool")
+            raise
+
+    actorpool_thread = threading.Thread(target=run_pool, name="actorpool")
+
+    num_updates = max(1, -(-flags.total_steps // steps_per_update))  # ceil
+    done_so_far = stats.get("step", 0) // (flags.unroll_length * flags.batch_size)
+    update_counter = {"mutex": threading.Lock(), "done": done_so_far}
+
+    learner_threads = [
+        threading.Thread(
+            target=learn,
+            name=f"learner-{i}",
+            args=(flags, learner_queue, model, flat_param, flat_grad, actor_flat,
+                  optimizer, scheduler, stats, plogger, reducer, num_updates,
+                  update_counter),
+        )
+        # Collectives must stay ordered: one learner thread under DP.
+        for i in range(1 if world_size > 1 else flags.num_learner_threads)
+    ]
+    inference_threads = [
+        threading.Thread(
+            target=inference,
+            name=f"inference-{i}",
+            args=(flags, inference_batcher, actor_model),
+        )
+        for i in range(flags.num_inference_threads)
+    ]
+
+    actorpool_thread.start()
+    for t in learner_threads + inference_threads:
+        t.start()
+
+    def checkpoint():
+        if flags.disable_checkpoint or not is_leader:
+            return
+        logging.info("Saving checkpoint to %s", checkpointpath)
+        torch.save(
+            {
+                "model_state_dict": model.state_dict(),
+                "optimizer_state_dict": optimizer.state_dict(),
+                "scheduler_state_dict": scheduler.state_dict(),
+                "stats": stats,
+                "flags": vars(flags),
+            },
+            checkpointpath,
+        )
+
+    def format_value(x):
+        return f"{x:1.5}" if isinstance(x, float) else str(x)
+
+    timer = timeit.default_timer
+    try:
+        last_checkpoint_time = timer()
+        while update_counter["done"] < num_updates and any(
+            t.is_alive() for t in learner_threads
+        ):
+            start_time = timer()
+            start_step = stats.get("step", 0)
+            time.sleep(5)
+            end_step = stats.get("step", 0)
+
+            if timer() - last_checkpoint_time > 10 * 60:
+                checkpoint()
+                last_checkpoint_time = timer()
+
+            if is_leader:
+                logging.info(
+                    "Step %i @ %.1f SPS (x%d ranks). Inference batcher size: %i."
+                    " Learner queue size: %i. Other stats: (%s)",
+                    end_step,
+                    (end_step - start_step) / (timer() - start_time),
+                    world_size,
+                    inference_batcher.size(),
+                    learner_queue.size(),
+                    ", ".join(
+                        f"{key} = {format_value(value)}"
+                        for key, value in stats.items()
+                        if key not in ("episode_returns",)
+                    ),
+                )
+    except KeyboardInterrupt:
+        pass
+    finally:
+        inference_batcher.close()
+        learner_queue.close()
+        for t in learner_threads + inference_threads:
+            t.join(timeout=10)
+        actorpool_thread.join(timeout=10)
+        checkpoint()
+        if plogger is not None:
+            plogger.close()
+
+    logging.info("Rank %d done after %d updates.", rank, update_counter["done"])
+
+
+def test(flags):
+    """Greedy-policy evaluation. The reference leaves polybeast test
+    unimplemented (ref: polybeast_learner.py:596-597); here we delegate to
+    the monobeast-style local-env evaluator."""
+    from torchbeast_amd import monobeast
+
+    return monobeast.test(flags)
+
+
+def main(flags):
+    if not hasattr(flags, "checkpoint_exists_ok"):
+        flags.checkpoint_exists_ok = True
+    if flags.write_profiler_trace:
+        logging.info("Running with profiler.")
+        with torch.profiler.profile(
+            activities=[
+                torch.profiler.ProfilerActivity.CPU,
+                torch.profiler.ProfilerActivity.CUDA,
+            ]
+        ) as prof:
+            if flags.mode == "train":
+                train(flags)
+            else:
+                test(flags)
+        filename = "chrome-%s.trace" % time.strftime("%Y%m%d-%H%M%S")
+        logging.info("Writing profiler trace to '%s'", filename)
+        prof.export_chrome_trace(filename)
+    else:
+        if flags.mode == "train":
+            train(flags)
+        else:
+            test(flags)
+
+
+if __name__ == "__main__":
+    main(parser.parse_args())

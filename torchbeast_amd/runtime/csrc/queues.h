@@ -1,0 +1,378 @@
+// Batching queues for the actor-learner runtime.
+//
+// Capability parity with the reference's BatchingQueue / DynamicBatcher
+// (ref: src/cc/actorpool.cc:72-340), redesigned for MI355X:
+// - dequeue/batch assembly concatenates into HIP-*pinned* host tensors
+//   (cat_out into a pinned destination) so the learner/inference H2D copy
+//   is a true DMA on a side stream, instead of pageable torch::cat output.
+// - one generic bounded MPMC queue underlies both the learner rollout queue
+//   and the inference request queue.
+
+#pragma once
+
+#include <torch/extension.h>
+
+#include <chrono>
+#include <condition_variable>
+#include <deque>
+#include <future>
+#include <memory>
+#include <mutex>
+#include <optional>
+#include <stdexcept>
+#include <vector>
+
+#include "nest.h"
+
+namespace tbruntime {
+
+using TensorNest = Nest<torch::Tensor>;
+
+class ClosedQueue : public std::runtime_error {
+ public:
+  using std::runtime_error::runtime_error;
+};
+
+class AsyncError : public std::runtime_error {
+ public:
+  using std::runtime_error::runtime_error;
+};
+
+inline bool pinned_memory_wanted() {
+  static const bool wanted = [] {
+    if (std::getenv("TBAMD_NO_PIN")) return false;
+    return torch::cuda::is_available();
+  }();
+  return wanted;
+}
+
+// Concatenate leaves along `dim` into a pinned destination when a GPU is
+// present (so the later .to(device, non_blocking=True) is DMA).
+inline torch::Tensor cat_pinned(const std::vector<torch::Tensor>& tensors,
+                                int64_t dim) {
+  TORCH_CHECK(!tensors.empty(), "cat_pinned: empty tensor list");
+  if (!pinned_memory_wanted() || tensors[0].is_cuda()) {
+    return torch::cat(tensors, dim);
+  }
+  auto shape = tensors[0].sizes().vec();
+  int64_t total = 0;
+  for (const auto& t : tensors) total += t.size(dim);
+  shape[dim] = total;
+  torch::Tensor out = torch::empty(
+      shape, tensors[0].options().pinned_memory(true));
+  torch::cat_out(out, tensors, dim);
+  return out;
+}
+
+inline TensorNest batch_nests(const std::vector<const TensorNest*>& nests,
+                              int64_t batch_dim) {
+  return TensorNest::apply_columns(
+      nests, [batch_dim](const std::vector<torch::Tensor>& column) {
+        return cat_pinned(column, batch_dim);
+      });
+}
+
+// ---------------------------------------------------------------------------
+// Generic bounded MPMC queue.
+// ---------------------------------------------------------------------------
+
+template <typename Item>
+class BoundedQueue {
+ public:
+  explicit BoundedQueue(std::optional<int64_t> max_size = std::nullopt)
+      : max_size_(max_size) {}
+
+  void enqueue(Item item) {
+    {
+      std::unique_lock<std::mutex> lock(mu_);
+      not_full_.wait(lock, [this] {
+        return closed_ || !max_size_ ||
+               static_cast<int64_t>(items_.size()) < *max_size_;
+      });
+      if (closed_) throw ClosedQueue("enqueue to closed queue");
+      items_.push_back(std::move(item));
+    }
+    not_empty_.notify_one();
+  }
+
+  // Block until at least min_n items (or timeout with >=1, or close).
+  // Returns up to max_n items. Throws ClosedQueue when closed and drained.
+  std::vector<Item> dequeue_many(int64_t min_n, int64_t max_n,
+                                 std::optional<std::chrono::milliseconds>
+                                     timeout = std::nullopt) {
+    std::unique_lock<std::mutex> lock(mu_);
+    auto have_min = [this, min_n] {
+      return closed_ || static_cast<int64_t>(items_.size()) >= min_n;
+    };
+    if (timeout) {
+      // After the deadline, settle for any non-empty prefix.
+      if (!not_empty_.wait_for(lock, *timeout, have_min)) {
+        not_empty_.wait(lock, [this] { return closed_ || !items_.empty(); });
+      }
+    } else {
+      not_empty_.wait(lock, have_min);
+    }
+    if (items_.empty()) {
+      // Only reachable when closed.
+      throw ClosedQueue("queue is closed");
+    }
+    int64_t n = std::min<int64_t>(items_.size(), max_n);
+    std::vector<Item> out;
+    out.reserve(n);
+    for (int64_t i = 0; i < n; ++i) {
+      out.push_back(std::move(items_.front()));
+      items_.pop_front();
+    }
+    lock.unlock();
+    not_full_.notify_all();
+    return out;
+  }
+
+  int64_t size() const {
+    std::lock_guard<std::mutex> lock(mu_);
+    return items_.size();
+  }
+
+  void close() {
+    {
+      std::lock_guard<std::mutex> lock(mu_);
+      if (closed_) throw ClosedQueue("queue was already closed");
+      closed_ = true;
+    }
+    not_empty_.notify_all();
+    not_full_.notify_all();
+  }
+
+  bool is_closed() const {
+    std::lock_guard<std::mutex> lock(mu_);
+    return closed_;
+  }
+
+ private:
+  mutable std::mutex mu_;
+  std::condition_variable not_empty_;
+  std::condition_variable not_full_;
+  std::deque<Item> items_;
+  std::optional<int64_t> max_size_;
+  bool closed_ = false;
+};
+
+// ---------------------------------------------------------------------------
+// Learner-facing queue of rollouts, batched along batch_dim on dequeue.
+// ---------------------------------------------------------------------------
+
+class BatchingQueue {
+ public:
+  BatchingQueue(int64_t batch_dim = 0,
+                std::optional<int64_t> minimum_batch_size = std::nullopt,
+                std::optional<int64_t> maximum_batch_size = std::nullopt,
+                std::optional<int64_t> timeout_ms = std::nullopt,
+                bool check_inputs = true,
+                std::optional<int64_t> maximum_queue_size = std::nullopt)
+      : batch_dim_(batch_dim),
+        min_batch_size_(minimum_batch_size ? *minimum_batch_size : 1),
+        max_batch_size_(maximum_batch_size
+                            ? *maximum_batch_size
+                            : std::numeric_limits<int64_t>::max()),
+        timeout_(timeout_ms
+                     ? std::optional<std::chrono::milliseconds>(
+                           std::chrono::milliseconds(*timeout_ms))
+                     : std::nullopt),
+        check_inputs_(check_inputs),
+        queue_(maximum_queue_size) {
+    if (min_batch_size_ < 1) {
+      throw std::invalid_argument("Min batch size must be >= 1");
+    }
+    if (max_batch_size_ < min_batch_size_) {
+      throw std::invalid_argument(
+          "Max batch size must be >= min batch size");
+    }
+    if (maximum_queue_size && *maximum_queue_size < 1) {
+      throw std::invalid_argument("Max queue size must be >= 1");
+    }
+  }
+
+  int64_t batch_dim() const { return batch_dim_; }
+
+  void enqueue(TensorNest item) {
+    if (check_inputs_) {
+      if (item.empty()) throw std::invalid_argument("Empty input");
+      item.for_each([this](const torch::Tensor& t) {
+        if (t.dim() <= batch_dim_) {
+          throw std::invalid_argument(
+              "Enqueued tensors must have more than batch_dim dims");
+        }
+      });
+    }
+    queue_.enqueue(std::move(item));
+  }
+
+  // One batched nest (cat along batch_dim) + the number of rollouts in it.
+  std::pair<TensorNest, int64_t> dequeue_many() {
+    std::vector<TensorNest> items =
+        queue_.dequeue_many(min_batch_size_, max_batch_size_, timeout_);
+    std::vector<const TensorNest*> ptrs;
+    ptrs.reserve(items.size());
+    for (const auto& n : items) ptrs.push_back(&n);
+    return {batch_nests(ptrs, batch_dim_), static_cast<int64_t>(items.size())};
+  }
+
+  int64_t size() const { return queue_.size(); }
+  void close() { queue_.close(); }
+  bool is_closed() const { return queue_.is_closed(); }
+
+ private:
+  const int64_t batch_dim_;
+  const int64_t min_batch_size_;
+  const int64_t max_batch_size_;
+  const std::optional<std::chrono::milliseconds> timeout_;
+  const bool check_inputs_;
+  BoundedQueue<TensorNest> queue_;
+};
+
+// ---------------------------------------------------------------------------
+// DynamicBatcher: many blocking compute() callers -> one batched inference.
+// ---------------------------------------------------------------------------
+
+class DynamicBatcher {
+ public:
+  struct Request {
+    TensorNest inputs;
+    int64_t batch_size;  // along batch_dim
+    std::shared_ptr<std::promise<TensorNest>> promise;
+  };
+
+  class Batch {
+   public:
+    Batch(int64_t batch_dim, std::vector<Request> requests, bool check_outputs)
+        : batch_dim_(batch_dim),
+          requests_(std::move(requests)),
+          check_outputs_(check_outputs) {}
+
+    ~Batch() {
+      if (!fulfilled_) {
+        auto eptr = std::make_exception_ptr(
+            AsyncError("Batch destroyed before set_outputs was called"));
+        for (auto& req : requests_) req.promise->set_exception(eptr);
+      }
+    }
+
+    int64_t size() const {
+      int64_t total = 0;
+      for (const auto& r : requests_) total += r.batch_size;
+      return total;
+    }
+
+    TensorNest get_inputs() {
+      std::vector<const TensorNest*> ptrs;
+      ptrs.reserve(requests_.size());
+      for (const auto& r : requests_) ptrs.push_back(&r.inputs);
+      return batch_nests(ptrs, batch_dim_);
+    }
+
+    void set_outputs(TensorNest outputs) {
+      if (fulfilled_) {
+        throw std::runtime_error("set_outputs called twice");
+      }
+      if (check_outputs_) {
+        const int64_t expected = size();
+        outputs.for_each([this, expected](const torch::Tensor& t) {
+          if (t.dim() <= batch_dim_) {
+            throw std::invalid_argument(
+                "With batch_dim == " + std::to_string(batch_dim_) +
+                ", output shape must have at least " +
+                std::to_string(batch_dim_ + 1) + " dims but got " +
+                std::to_string(t.dim()));
+          }
+          if (t.size(batch_dim_) != expected) {
+            throw std::invalid_argument(
+                "Output shape must have the same batch dimension as the "
+                "input batch size. Expected: " + std::to_string(expected) +
+                ". Observed: " + std::to_string(t.size(batch_dim_)));
+          }
+        });
+      }
+      int64_t offset = 0;
+      for (auto& req : requests_) {
+        TensorNest slice =
+            outputs.map([this, offset, &req](const torch::Tensor& t) {
+              return t.narrow(batch_dim_, offset, req.batch_size);
+            });
+        req.promise->set_value(std::move(slice));
+        offset += req.batch_size;
+      }
+      fulfilled_ = true;
+    }
+
+   private:
+    const int64_t batch_dim_;
+    std::vector<Request> requests_;
+    const bool check_outputs_;
+    bool fulfilled_ = false;
+  };
+
+  DynamicBatcher(int64_t batch_dim = 0,
+                 std::optional<int64_t> minimum_batch_size = std::nullopt,
+                 std::optional<int64_t> maximum_batch_size = std::nullopt,
+                 std::optional<int64_t> timeout_ms = std::nullopt,
+                 bool check_outputs = true)
+      : batch_dim_(batch_dim),
+        min_batch_size_(minimum_batch_size ? *minimum_batch_size : 1),
+        max_batch_size_(maximum_batch_size
+                            ? *maximum_batch_size
+                            : std::numeric_limits<int64_t>::max()),
+        timeout_(timeout_ms
+                     ? std::optional<std::chrono::milliseconds>(
+                           std::chrono::milliseconds(*timeout_ms))
+                     : std::nullopt),
+        check_outputs_(check_outputs),
+        queue_(std::nullopt) {}
+
+  // Called by actor threads; blocks until the consumer sets outputs.
+  TensorNest compute(TensorNest inputs) {
+    if (inputs.empty()) {
+      throw std::invalid_argument("compute() on empty nest");
+    }
+    int64_t batch_size = -1;
+    inputs.for_each([this, &batch_size](const torch::Tensor& t) {
+      if (t.dim() <= batch_dim_) {
+        throw std::invalid_argument("Input needs more dims than batch_dim");
+      }
+      if (batch_size < 0) {
+        batch_size = t.size(batch_dim_);
+      } else if (t.size(batch_dim_) != batch_size) {
+        throw std::invalid_argument(
+            "Input tensors disagree on the batch dimension");
+      }
+    });
+    auto promise = std::make_shared<std::promise<TensorNest>>();
+    std::future<TensorNest> future = promise->get_future();
+    queue_.enqueue(Request{std::move(inputs), batch_size, std::move(promise)});
+    if (future.wait_for(std::chrono::minutes(10)) ==
+        std::future_status::timeout) {
+      throw AsyncError("compute() timed out after 10 minutes");
+    }
+    return future.get();  // Rethrows AsyncError from a dropped batch.
+  }
+
+  std::shared_ptr<Batch> get_batch() {
+    std::vector<Request> requests =
+        queue_.dequeue_many(min_batch_size_, max_batch_size_, timeout_);
+    return std::make_shared<Batch>(batch_dim_, std::move(requests),
+                                   check_outputs_);
+  }
+
+  int64_t size() const { return queue_.size(); }
+  void close() { queue_.close(); }
+  bool is_closed() const { return queue_.is_closed(); }
+
+ private:
+  const int64_t batch_dim_;
+  const int64_t min_batch_size_;
+  const int64_t max_batch_size_;
+  const std::optional<std::chrono::milliseconds> timeout_;
+  const bool check_outputs_;
+  BoundedQueue<Request> queue_;
+};
+
+}  // namespace tbruntime
