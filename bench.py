@@ -1,0 +1,253 @@
+"""Flagship benchmark: polybeast IMPALA training SPS on MI355X.
+
+Measures the BASELINE.json metric — env-steps/sec (SPS), whole node,
+AtariNet 84x84x4, unroll=80 — on synthetic Atari-shaped frames with
+random-init weights, running the full actor-learner pipeline per GPU:
+native synthetic envs -> C++ ActorPool -> DynamicBatcher -> batched GPU
+inference -> rollout BatchingQueue -> learner step (V-trace + fused IMPALA
+loss + fused clip/RMSProp), with a flat-gradient RCCL all-reduce across
+ranks when N > 1.
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  # N>1: python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+  #        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+A "step" is one learner update consuming unroll_length*batch_size env steps.
+Rank 0 prints ONE JSON line with the whole-job aggregate SPS.
+"""
+
+import argparse
+import json
+import os
+import threading
+import timeit
+
+import torch
+
+import torchbeast_amd.polybeast_learner as pbl
+from torchbeast_amd import runtime
+from torchbeast_amd.core import vtrace
+from torchbeast_amd.ops import functional as tbops
+from torchbeast_amd.parallel import ddp as tbddp
+from torchbeast_amd.parallel import flat as tbflat
+
+METRIC = (
+    "env-steps/sec (SPS) whole node, AtariNet 84×84×4 unroll=80 "
+    "at 1/2/4/8 MI355X"
+)
+
+
+def parse_args():
+    p = argparse.ArgumentParser(description="torchbeast_amd flagship benchmark")
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch_size", type=int, default=32)
+    p.add_argument("--unroll_length", type=int, default=80)
+    p.add_argument("--actors", type=int, default=64,
+                   help="Env streams per GPU.")
+    p.add_argument("--model", default="shallow", choices=["shallow", "deep"])
+    p.add_argument("--use_lstm", action="store_true")
+    p.add_argument("--frame", default="4x84x84",
+                   help="Synthetic frame shape CxHxW.")
+    p.add_argument("--num_actions", type=int, default=6)
+    p.add_argument("--num_inference_threads", type=int, default=2)
+    p.add_argument("--inference_max_batch_size", type=int, default=512)
+    p.add_argument("--inference_timeout_ms", type=int, default=5)
+    p.add_argument("--episode_length", type=int, default=1000)
+    return p.parse_args()
+
+
+def learner_step(flags, batch_tensors, model, optimizer, scheduler, reducer,
+                 actor_flat, flat_param):
+    env_outputs = pbl.EnvOutput._make(batch_tensors[:5])
+    actor_outputs = pbl.AgentOutput._make(batch_tensors[5:8])
+    initial_agent_state = batch_tensors[8:]
+
+    learner_outputs, _ = model(
+        dict(frame=env_outputs.frame, reward=env_outputs.rewards,
+             done=env_outputs.done),
+        initial_agent_state,
+    )
+    learner_outputs = pbl.AgentOutput._make(pbl._as_agent_output(learner_outputs))
+    bootstrap_value = learner_outputs.baseline[-1]
+
+    env_outputs = pbl.EnvOutput._make(t[1:] for t in env_outputs)
+    actor_outputs = pbl.AgentOutput._make(t[1:] for t in actor_outputs)
+    learner_outputs = pbl.AgentOutput._make(t[:-1] for t in learner_outputs)
+
+    clipped_rewards = torch.clamp(env_outputs.rewards, -1, 1)
+    discounts = (~env_outputs.done).float() * flags.discounting
+
+    vtr = vtrace.from_logits(
+        behavior_policy_logits=actor_outputs.policy_logits,
+        target_policy_logits=learner_outputs.policy_logits,
+        actions=actor_outputs.action,
+        discounts=discounts,
+        rewards=clipped_rewards,
+        values=learner_outputs.baseline,
+        bootstrap_value=bootstrap_value,
+    )
+    pg_loss, baseline_loss, entropy_loss = tbops.fused_impala_loss(
+        learner_outputs.policy_logits,
+        learner_outputs.baseline,
+        actor_outputs.action,
+        vtr.pg_advantages,
+        vtr.vs,
+    )
+    total_loss = (pg_loss + flags.baseline_cost * baseline_loss
+                  + flags.entropy_cost * entropy_loss)
+
+    optimizer.zero_grad()
+    total_loss.backward()
+    reducer.reduce()
+    optimizer.step()
+    scheduler.step()
+    with torch.no_grad():
+        actor_flat.copy_(flat_param)
+    return total_loss
+
+
+def main():
+    args = parse_args()
+    rank, world_size, local_rank = tbddp.maybe_init_distributed()
+
+    use_cuda = torch.cuda.is_available()
+    device = torch.device("cuda", local_rank) if use_cuda else torch.device("cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+
+    # Reuse the polybeast flag namespace for model/loss hyperparameters.
+    flags = pbl.parser.parse_args([])
+    flags.env = f"synthetic:{args.frame}:{args.num_actions}:{args.episode_length}"
+    flags.model = args.model
+    flags.use_lstm = args.use_lstm
+    flags.num_actions = args.num_actions
+    flags.batch_size = args.batch_size
+    flags.unroll_length = args.unroll_length
+    flags.learner_device = device
+    flags.actor_device = device
+
+    T, B = args.unroll_length, args.batch_size
+
+    model = pbl.create_model(flags).to(device)
+    actor_model = pbl.create_model(flags).to(device)
+    flat_param = tbflat.flatten_parameters(model)
+    flat_grad = tbflat.attach_flat_grads(model)
+    actor_flat = tbflat.flatten_parameters(actor_model)
+    tbddp.broadcast_flat(flat_param)
+    with torch.no_grad():
+        actor_flat.copy_(flat_param)
+
+    optimizer = tbflat.FusedRMSProp(
+        flat_param, flat_grad, lr=flags.learning_rate, alpha=flags.alpha,
+        eps=flags.epsilon, clip_norm=flags.grad_norm_clipping,
+    )
+    total_env_steps = (args.warmup + args.steps) * T * B * world_size * 10
+    scheduler = tbflat.LinearLR(optimizer, T * B * world_size, total_env_steps)
+    reducer = tbddp.GradAllReducer(flat_grad, world_size)
+
+    learner_queue = runtime.BatchingQueue(
+        batch_dim=1, minimum_batch_size=B, maximum_batch_size=B,
+        maximum_queue_size=B,
+    )
+    inference_batcher = runtime.DynamicBatcher(
+        batch_dim=1, minimum_batch_size=1,
+        maximum_batch_size=args.inference_max_batch_size,
+        timeout_ms=args.inference_timeout_ms,
+    )
+    addresses = [flags.env] * args.actors
+    initial_agent_state = tuple(
+        t.cpu() for t in model.initial_state(batch_size=1)
+    )
+    pool = runtime.ActorPool(
+        unroll_length=T, learner_queue=learner_queue,
+        inference_batcher=inference_batcher,
+        env_server_addresses=addresses,
+        initial_agent_state=initial_agent_state,
+    )
+    pool_thread = threading.Thread(target=pool.run, daemon=True)
+    pool_thread.start()
+
+    inference_threads = [
+        threading.Thread(target=pbl.inference,
+                         args=(flags, inference_batcher, actor_model),
+                         daemon=True)
+        for _ in range(args.num_inference_threads)
+    ]
+    for t in inference_threads:
+        t.start()
+
+    def get_batch(it):
+        tensors = runtime._tbruntime.flatten(next(it))
+        return tuple(t.to(device, non_blocking=True) for t in tensors)
+
+    queue_iter = iter(learner_queue)
+
+    # Warmup (fills the pipeline, compiles/caches kernels).
+    for _ in range(args.warmup):
+        learner_step(flags, get_batch(queue_iter), model, optimizer, scheduler,
+                     reducer, actor_flat, flat_param)
+
+    if world_size > 1:
+        torch.distributed.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    start = timeit.default_timer()
+
+    for _ in range(args.steps):
+        learner_step(flags, get_batch(queue_iter), model, optimizer, scheduler,
+                     reducer, actor_flat, flat_param)
+
+    if use_cuda:
+        torch.cuda.synchronize()
+    if world_size > 1:
+        torch.distributed.barrier()
+    elapsed = timeit.default_timer() - start
+
+    # MAX elapsed over ranks (barriers make local elapsed ≈ max already, but
+    # reduce explicitly for correctness).
+    if world_size > 1:
+        e = torch.tensor([elapsed], device=device if use_cuda else None)
+        torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    inference_batcher.close()
+    learner_queue.close()
+
+    if rank == 0:
+        total_env_steps = args.steps * T * B * world_size
+        sps = total_env_steps / elapsed
+        result = {
+            "metric": METRIC,
+            "value": sps,
+            "unit": "env_steps/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "AtariNet" if args.model == "shallow" else "IMPALA-ResNet",
+                "global_batch": B * world_size,
+                "seq_len": T,
+                "parallelism": f"dp{world_size}",
+                "actors_per_gpu": args.actors,
+                "use_lstm": args.use_lstm,
+                "frame": args.frame,
+            },
+        }
+        print(json.dumps(result))
+
+    if world_size > 1:
+        torch.distributed.destroy_process_group()
+    os._exit(0)  # Actor/inference threads are daemonic; skip teardown races.
+
+
+if __name__ == "__main__":
+    main()

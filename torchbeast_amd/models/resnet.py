@@ -88,25 +88,23 @@ class ResNet(nn.Module):
         core_input = torch.cat([x, clipped_reward], dim=-1)
 
         if self.use_lstm:
+            from torchbeast_amd.ops import functional as tbf
+
             core_input = core_input.view(T, B, -1)
             notdone = (~done).float()
-            outputs = []
-            for step_input, nd in zip(core_input.unbind(), notdone.unbind()):
-                nd = nd.view(1, -1, 1)
-                core_state = tuple(nd * s for s in core_state)
-                out, core_state = self.core(step_input.unsqueeze(0), core_state)
-                outputs.append(out)
-            core_output = torch.flatten(torch.cat(outputs), 0, 1)
+            core_output_seq, core_state = tbf.lstm_unroll(
+                self.core, core_input, notdone, core_state
+            )
+            core_output = torch.flatten(core_output_seq, 0, 1)
         else:
             core_output = core_input
 
         policy_logits = self.policy(core_output)
         baseline = self.baseline(core_output)
 
-        if self.training:
-            action = torch.multinomial(F.softmax(policy_logits, dim=1), num_samples=1)
-        else:
-            action = torch.argmax(policy_logits, dim=1, keepdim=True)
+        from torchbeast_amd.ops import functional as tbf
+
+        action = tbf.policy_sample(policy_logits, greedy=not self.training)
 
         policy_logits = policy_logits.view(T, B, self.num_actions)
         baseline = baseline.view(T, B)

@@ -162,33 +162,39 @@ class _LstmUnroll(torch.autograd.Function):
     def forward(ctx, x, notdone, h0, c0, *flat_weights):
         ext = _ext_for(x, "lstm_unroll_fwd")
         num_layers = len(flat_weights) // 4
-        out, hT, cT, stash = ext.lstm_unroll_fwd(
+        results = ext.lstm_unroll_fwd(
             x.contiguous(),
             notdone.contiguous(),
             h0.contiguous(),
             c0.contiguous(),
-            list(flat_weights),
+            [w.contiguous() for w in flat_weights],
         )
-        ctx.save_for_backward(x, notdone, h0, c0, stash, *flat_weights)
+        out, hT, cT = results[:3]
+        stash = results[3:]  # Per layer: input, gates, hm, cm, c.
+        ctx.save_for_backward(notdone, h0, c0, *flat_weights, *stash)
         ctx.num_layers = num_layers
         return out, hT, cT
 
     @staticmethod
     def backward(ctx, d_out, d_hT, d_cT):
-        x, notdone, h0, c0, stash, *flat_weights = ctx.saved_tensors
+        saved = ctx.saved_tensors
+        notdone, h0, c0 = saved[:3]
+        nw = ctx.num_layers * 4
+        flat_weights = list(saved[3 : 3 + nw])
+        stash = list(saved[3 + nw :])
         ext = ops_mod.require_ext()
         grads = ext.lstm_unroll_bwd(
-            x,
             notdone,
             h0,
             c0,
+            flat_weights,
             stash,
-            list(flat_weights),
             d_out.contiguous(),
             d_hT.contiguous(),
             d_cT.contiguous(),
         )
-        d_x, d_h0, d_c0, *d_weights = grads
+        d_x, d_h0, d_c0 = grads[:3]
+        d_weights = grads[3:]
         return (d_x, None, d_h0, d_c0, *d_weights)
 
 
