@@ -242,7 +242,7 @@ def main():
                 "frame": args.frame,
             },
         }
-        print(json.dumps(result))
+        print(json.dumps(result), flush=True)
 
     if world_size > 1:
         torch.distributed.destroy_process_group()
