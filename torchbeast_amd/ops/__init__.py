@@ -28,23 +28,32 @@ _tbops = None
 _tbops_error = None
 
 
+def _raise_if_gpu():
+    if torch.cuda.is_available() and not os.environ.get("TBAMD_ALLOW_EAGER"):
+        raise ImportError(
+            "torchbeast_amd HIP extension (_tbops) failed to import but a GPU "
+            "is present; build it with `python setup.py build_ext --inplace` "
+            "(PYTORCH_ROCM_ARCH=gfx950). Set TBAMD_ALLOW_EAGER=1 to run on "
+            f"stock PyTorch ops anyway. Original error: {_tbops_error!r}"
+        ) from _tbops_error
+
+
 def _load():
     global _tbops, _tbops_error
-    if _tbops is not None or _tbops_error is not None:
+    if _tbops is not None:
         return _tbops
+    if _tbops_error is not None:
+        _raise_if_gpu()
+        return None
     try:
-        from torchbeast_amd.ops import _tbops as ext  # noqa: F401
+        # importlib, NOT `from torchbeast_amd.ops import _tbops`: this
+        # module's own `_tbops = None` global would shadow the submodule.
+        import importlib
 
-        _tbops = ext
-    except ImportError as e:
+        _tbops = importlib.import_module("torchbeast_amd.ops._tbops")
+    except Exception as e:  # ImportError, OSError (dlopen), RuntimeError (HIP)
         _tbops_error = e
-        if torch.cuda.is_available() and not os.environ.get("TBAMD_ALLOW_EAGER"):
-            raise ImportError(
-                "torchbeast_amd HIP extension (_tbops) is not built but a GPU "
-                "is present; build it with `python setup.py build_ext --inplace` "
-                "(PYTORCH_ROCM_ARCH=gfx950). Set TBAMD_ALLOW_EAGER=1 to run "
-                "on stock PyTorch ops anyway."
-            ) from e
+        _raise_if_gpu()
     return _tbops
 
 
@@ -60,7 +69,8 @@ def require_ext():
     ext = _load()
     if ext is None:
         raise ImportError(
-            "torchbeast_amd._tbops is required here but not built"
+            "torchbeast_amd._tbops is required here but not built "
+            f"(original error: {_tbops_error!r})"
         ) from _tbops_error
     return ext
 
