@@ -20,6 +20,7 @@ Rank 0 prints ONE JSON line with the whole-job aggregate SPS.
 import argparse
 import json
 import os
+import sys
 import threading
 import timeit
 
@@ -190,6 +191,12 @@ def main():
         learner_step(flags, get_batch(queue_iter), model, optimizer, scheduler,
                      reducer, actor_flat, flat_param)
 
+    timings = None
+    if os.environ.get("TBAMD_BENCH_TIMINGS"):
+        from torchbeast_amd.core.prof import Timings
+
+        timings = Timings()
+
     if world_size > 1:
         torch.distributed.barrier()
     if use_cuda:
@@ -197,8 +204,18 @@ def main():
     start = timeit.default_timer()
 
     for _ in range(args.steps):
-        learner_step(flags, get_batch(queue_iter), model, optimizer, scheduler,
-                     reducer, actor_flat, flat_param)
+        if timings is None:
+            learner_step(flags, get_batch(queue_iter), model, optimizer,
+                         scheduler, reducer, actor_flat, flat_param)
+        else:
+            timings.reset()
+            batch = get_batch(queue_iter)
+            torch.cuda.synchronize()
+            timings.time("get_batch")
+            learner_step(flags, batch, model, optimizer, scheduler, reducer,
+                         actor_flat, flat_param)
+            torch.cuda.synchronize()
+            timings.time("learn")
 
     if use_cuda:
         torch.cuda.synchronize()
@@ -212,6 +229,10 @@ def main():
         e = torch.tensor([elapsed], device=device if use_cuda else None)
         torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(e.item())
+
+    if timings is not None and rank == 0:
+        print("bench timings:" + timings.summary(), file=sys.stderr)
+        print(f"inference stats: {pbl.INFERENCE_STATS}", file=sys.stderr)
 
     inference_batcher.close()
     learner_queue.close()
@@ -246,6 +267,11 @@ def main():
 
     if world_size > 1:
         torch.distributed.destroy_process_group()
+    if os.environ.get("TBAMD_CLEAN_EXIT"):
+        # Orderly exit (lets rocprofv3 flush its results); threads are
+        # daemonic and unwind via the closed queues.
+        pool_thread.join(timeout=5)
+        sys.exit(0)
     os._exit(0)  # Actor/inference threads are daemonic; skip teardown races.
 
 

@@ -95,10 +95,15 @@ def _as_agent_output(outputs):
     return tuple(outputs)
 
 
+INFERENCE_STATS = {"batches": 0, "env_steps": 0}
+
+
 def inference(flags, inference_batcher, model, lock=threading.Lock()):  # noqa: B008
     """Consume inference batches: one batched behavior-model forward each."""
     with torch.no_grad():
         for batch in inference_batcher:
+            INFERENCE_STATS["batches"] += 1
+            INFERENCE_STATS["env_steps"] += batch.size()
             batched_env_outputs, agent_state = batch.get_inputs()
             frame, reward, done, *_ = batched_env_outputs
             frame = frame.to(flags.actor_device, non_blocking=True)
