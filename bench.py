@@ -54,6 +54,7 @@ def parse_args():
                    help="Synthetic frame shape CxHxW.")
     p.add_argument("--num_actions", type=int, default=6)
     p.add_argument("--num_inference_threads", type=int, default=2)
+    p.add_argument("--inference_min_batch_size", type=int, default=1)
     p.add_argument("--inference_max_batch_size", type=int, default=512)
     p.add_argument("--inference_timeout_ms", type=int, default=5)
     p.add_argument("--episode_length", type=int, default=1000)
@@ -154,7 +155,7 @@ def main():
         maximum_queue_size=B,
     )
     inference_batcher = runtime.DynamicBatcher(
-        batch_dim=1, minimum_batch_size=1,
+        batch_dim=1, minimum_batch_size=args.inference_min_batch_size,
         maximum_batch_size=args.inference_max_batch_size,
         timeout_ms=args.inference_timeout_ms,
     )

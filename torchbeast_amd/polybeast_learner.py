@@ -62,6 +62,7 @@ def make_parser():
     parser.add_argument("--model", default="shallow", choices=["shallow", "deep"],
                         help="shallow=AtariNet (the headline bench model), "
                              "deep=IMPALA ResNet.")
+    parser.add_argument("--inference_min_batch_size", default=1, type=int)
     parser.add_argument("--inference_max_batch_size", default=512, type=int)
     parser.add_argument("--inference_timeout_ms", default=10, type=int)
     parser.add_argument("--write_profiler_trace", action="store_true")
@@ -97,10 +98,34 @@ def _as_agent_output(outputs):
 
 INFERENCE_STATS = {"batches": 0, "env_steps": 0}
 
+_inference_stream = None
+_inference_stream_lock = threading.Lock()
+
+
+def _get_inference_stream(device):
+    """One shared side stream for ALL inference work, so behavior-model
+    forwards never queue behind the learner's step on the default stream
+    (the reference serializes both on one device context; on MI355X the
+    side stream + pinned copies make inference latency independent of the
+    learner)."""
+    global _inference_stream
+    if device.type != "cuda":
+        return None
+    with _inference_stream_lock:
+        if _inference_stream is None:
+            _inference_stream = torch.cuda.Stream(device=device)
+    return _inference_stream
+
 
 def inference(flags, inference_batcher, model, lock=threading.Lock()):  # noqa: B008
     """Consume inference batches: one batched behavior-model forward each."""
-    with torch.no_grad():
+    stream = _get_inference_stream(flags.actor_device)
+    import contextlib
+
+    stream_ctx = (
+        torch.cuda.stream(stream) if stream is not None else contextlib.nullcontext()
+    )
+    with torch.no_grad(), stream_ctx:
         for batch in inference_batcher:
             INFERENCE_STATS["batches"] += 1
             INFERENCE_STATS["env_steps"] += batch.size()
@@ -116,9 +141,9 @@ def inference(flags, inference_batcher, model, lock=threading.Lock()):  # noqa: 
                 outputs, new_state = model(
                     dict(frame=frame, reward=reward, done=done), agent_state
                 )
-            outputs = _as_agent_output(outputs)
-            outputs = tuple(t.cpu() for t in outputs)
-            new_state = tuple(t.cpu() for t in new_state)
+                outputs = _as_agent_output(outputs)
+                outputs = tuple(t.cpu() for t in outputs)
+                new_state = tuple(t.cpu() for t in new_state)
             batch.set_outputs((outputs, new_state))
 
 
@@ -270,7 +295,7 @@ def train(flags):  # noqa: C901
     )
     inference_batcher = runtime.DynamicBatcher(
         batch_dim=1,
-        minimum_batch_size=1,
+        minimum_batch_size=flags.inference_min_batch_size,
         maximum_batch_size=flags.inference_max_batch_size,
         timeout_ms=flags.inference_timeout_ms,
         check_outputs=True,
