@@ -234,6 +234,8 @@ def main():
     if timings is not None and rank == 0:
         print("bench timings:" + timings.summary(), file=sys.stderr)
         print(f"inference stats: {pbl.INFERENCE_STATS}", file=sys.stderr)
+        print(f"learner_queue stats: {learner_queue.stats()}", file=sys.stderr)
+        print(f"batcher stats: {inference_batcher.stats()}", file=sys.stderr)
 
     inference_batcher.close()
     learner_queue.close()

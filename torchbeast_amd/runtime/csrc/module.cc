@@ -172,6 +172,7 @@ PYBIND11_MODULE(_tbruntime, m) {
       .def("enqueue", &BatchingQueue::enqueue, py::arg("nest"),
            py::call_guard<py::gil_scoped_release>())
       .def("size", &BatchingQueue::size)
+      .def("stats", &BatchingQueue::stats)
       .def("close", &BatchingQueue::close,
            py::call_guard<py::gil_scoped_release>())
       .def("is_closed", &BatchingQueue::is_closed)
@@ -211,6 +212,7 @@ PYBIND11_MODULE(_tbruntime, m) {
       .def("compute", &DynamicBatcher::compute, py::arg("inputs"),
            py::call_guard<py::gil_scoped_release>())
       .def("size", &DynamicBatcher::size)
+      .def("stats", &DynamicBatcher::stats)
       .def("close", &DynamicBatcher::close,
            py::call_guard<py::gil_scoped_release>())
       .def("is_closed", &DynamicBatcher::is_closed)

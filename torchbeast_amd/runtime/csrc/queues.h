@@ -12,10 +12,12 @@
 
 #include <torch/extension.h>
 
+#include <atomic>
 #include <chrono>
 #include <condition_variable>
 #include <deque>
 #include <future>
+#include <map>
 #include <memory>
 #include <mutex>
 #include <optional>
@@ -27,6 +29,22 @@
 namespace tbruntime {
 
 using TensorNest = Nest<torch::Tensor>;
+
+inline int64_t now_us() {
+  return std::chrono::duration_cast<std::chrono::microseconds>(
+             std::chrono::steady_clock::now().time_since_epoch())
+      .count();
+}
+
+// Lightweight perf counters (atomics; read via .stats() from Python).
+struct StageStats {
+  std::atomic<int64_t> n{0};
+  std::atomic<int64_t> sum_us{0};
+  void add(int64_t us) {
+    n.fetch_add(1, std::memory_order_relaxed);
+    sum_us.fetch_add(us, std::memory_order_relaxed);
+  }
+};
 
 class ClosedQueue : public std::runtime_error {
  public:
@@ -209,17 +227,33 @@ class BatchingQueue {
 
   // One batched nest (cat along batch_dim) + the number of rollouts in it.
   std::pair<TensorNest, int64_t> dequeue_many() {
+    const int64_t t0 = now_us();
     std::vector<TensorNest> items =
         queue_.dequeue_many(min_batch_size_, max_batch_size_, timeout_);
+    const int64_t t1 = now_us();
     std::vector<const TensorNest*> ptrs;
     ptrs.reserve(items.size());
     for (const auto& n : items) ptrs.push_back(&n);
-    return {batch_nests(ptrs, batch_dim_), static_cast<int64_t>(items.size())};
+    auto result = std::make_pair(batch_nests(ptrs, batch_dim_),
+                                 static_cast<int64_t>(items.size()));
+    wait_stats_.add(t1 - t0);
+    cat_stats_.add(now_us() - t1);
+    return result;
   }
 
   int64_t size() const { return queue_.size(); }
   void close() { queue_.close(); }
   bool is_closed() const { return queue_.is_closed(); }
+
+  std::map<std::string, double> stats() const {
+    std::map<std::string, double> out;
+    out["dequeues"] = wait_stats_.n.load();
+    if (wait_stats_.n > 0) {
+      out["avg_wait_ms"] = wait_stats_.sum_us / 1e3 / wait_stats_.n;
+      out["avg_cat_ms"] = cat_stats_.sum_us / 1e3 / cat_stats_.n;
+    }
+    return out;
+  }
 
  private:
   const int64_t batch_dim_;
@@ -228,6 +262,8 @@ class BatchingQueue {
   const std::optional<std::chrono::milliseconds> timeout_;
   const bool check_inputs_;
   BoundedQueue<TensorNest> queue_;
+  mutable StageStats wait_stats_;
+  mutable StageStats cat_stats_;
 };
 
 // ---------------------------------------------------------------------------
@@ -244,10 +280,13 @@ class DynamicBatcher {
 
   class Batch {
    public:
-    Batch(int64_t batch_dim, std::vector<Request> requests, bool check_outputs)
+    Batch(int64_t batch_dim, std::vector<Request> requests, bool check_outputs,
+          StageStats* service_stats = nullptr)
         : batch_dim_(batch_dim),
           requests_(std::move(requests)),
-          check_outputs_(check_outputs) {}
+          check_outputs_(check_outputs),
+          service_stats_(service_stats),
+          created_us_(now_us()) {}
 
     ~Batch() {
       if (!fulfilled_) {
@@ -302,12 +341,17 @@ class DynamicBatcher {
         offset += req.batch_size;
       }
       fulfilled_ = true;
+      if (service_stats_ != nullptr) {
+        service_stats_->add(now_us() - created_us_);
+      }
     }
 
    private:
     const int64_t batch_dim_;
     std::vector<Request> requests_;
     const bool check_outputs_;
+    StageStats* service_stats_;
+    const int64_t created_us_;
     bool fulfilled_ = false;
   };
 
@@ -347,24 +391,52 @@ class DynamicBatcher {
     });
     auto promise = std::make_shared<std::promise<TensorNest>>();
     std::future<TensorNest> future = promise->get_future();
+    const int64_t t0 = now_us();
     queue_.enqueue(Request{std::move(inputs), batch_size, std::move(promise)});
     if (future.wait_for(std::chrono::minutes(10)) ==
         std::future_status::timeout) {
       throw AsyncError("compute() timed out after 10 minutes");
     }
-    return future.get();  // Rethrows AsyncError from a dropped batch.
+    auto result = future.get();  // Rethrows AsyncError from a dropped batch.
+    roundtrip_stats_.add(now_us() - t0);
+    return result;
   }
 
   std::shared_ptr<Batch> get_batch() {
+    const int64_t t0 = now_us();
     std::vector<Request> requests =
         queue_.dequeue_many(min_batch_size_, max_batch_size_, timeout_);
+    formation_stats_.add(now_us() - t0);
+    int64_t total = 0;
+    for (const auto& r : requests) total += r.batch_size;
+    batch_size_stats_.add(total);
     return std::make_shared<Batch>(batch_dim_, std::move(requests),
-                                   check_outputs_);
+                                   check_outputs_, &service_stats_);
   }
 
   int64_t size() const { return queue_.size(); }
   void close() { queue_.close(); }
   bool is_closed() const { return queue_.is_closed(); }
+
+  std::map<std::string, double> stats() const {
+    std::map<std::string, double> out;
+    out["computes"] = roundtrip_stats_.n.load();
+    out["batches"] = batch_size_stats_.n.load();
+    if (roundtrip_stats_.n > 0) {
+      out["avg_roundtrip_ms"] =
+          roundtrip_stats_.sum_us / 1e3 / roundtrip_stats_.n;
+    }
+    if (batch_size_stats_.n > 0) {
+      out["avg_batch_size"] =
+          double(batch_size_stats_.sum_us) / batch_size_stats_.n;
+      out["avg_formation_wait_ms"] =
+          formation_stats_.sum_us / 1e3 / formation_stats_.n;
+    }
+    if (service_stats_.n > 0) {
+      out["avg_service_ms"] = service_stats_.sum_us / 1e3 / service_stats_.n;
+    }
+    return out;
+  }
 
  private:
   const int64_t batch_dim_;
@@ -373,6 +445,10 @@ class DynamicBatcher {
   const std::optional<std::chrono::milliseconds> timeout_;
   const bool check_outputs_;
   BoundedQueue<Request> queue_;
+  mutable StageStats roundtrip_stats_;
+  mutable StageStats formation_stats_;
+  mutable StageStats service_stats_;
+  mutable StageStats batch_size_stats_;  // sum_us field holds summed sizes
 };
 
 }  // namespace tbruntime
