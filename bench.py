@@ -58,6 +58,8 @@ def parse_args():
     p.add_argument("--inference_max_batch_size", type=int, default=512)
     p.add_argument("--inference_timeout_ms", type=int, default=5)
     p.add_argument("--episode_length", type=int, default=1000)
+    p.add_argument("--py_inference", action="store_true",
+                   help="Python inference threads instead of the C++ engine.")
     return p.parse_args()
 
 
@@ -172,14 +174,19 @@ def main():
     pool_thread = threading.Thread(target=pool.run, daemon=True)
     pool_thread.start()
 
-    inference_threads = [
-        threading.Thread(target=pbl.inference,
-                         args=(flags, inference_batcher, actor_model),
-                         daemon=True)
-        for _ in range(args.num_inference_threads)
-    ]
-    for t in inference_threads:
-        t.start()
+    inference_runner = None
+    if args.model == "shallow" and use_cuda and not args.py_inference:
+        inference_runner = pbl.make_inference_runner(
+            actor_model, inference_batcher
+        )
+        inference_runner.start(args.num_inference_threads)
+    else:
+        for _ in range(args.num_inference_threads):
+            threading.Thread(
+                target=pbl.inference,
+                args=(flags, inference_batcher, actor_model),
+                daemon=True,
+            ).start()
 
     def get_batch(it):
         tensors = runtime._tbruntime.flatten(next(it))
@@ -239,6 +246,8 @@ def main():
 
     inference_batcher.close()
     learner_queue.close()
+    if inference_runner is not None and os.environ.get("TBAMD_CLEAN_EXIT"):
+        inference_runner.stop()
 
     if rank == 0:
         total_env_steps = args.steps * T * B * world_size

@@ -19,10 +19,15 @@ os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 ROOT = os.path.dirname(os.path.abspath(__file__))
 
 ext_modules = [
-    CppExtension(
+    # CUDAExtension (ROCm) so the runtime can drive HIP streams/ATen-GPU
+    # ops from C++ (GIL-free inference engine).
+    CUDAExtension(
         name="torchbeast_amd.runtime._tbruntime",
         sources=["torchbeast_amd/runtime/csrc/module.cc"],
-        extra_compile_args=["-O3", "-std=c++17", "-pthread"],
+        extra_compile_args={
+            "cxx": ["-O3", "-std=c++17", "-pthread"],
+            "nvcc": ["-O3", "-std=c++17"],
+        },
     )
 ]
 

@@ -66,6 +66,10 @@ def make_parser():
     parser.add_argument("--inference_max_batch_size", default=512, type=int)
     parser.add_argument("--inference_timeout_ms", default=10, type=int)
     parser.add_argument("--write_profiler_trace", action="store_true")
+    parser.add_argument("--py_inference", action="store_true",
+                        help="Serve inference from Python threads instead of "
+                             "the C++ engine (always the case for --model "
+                             "deep or on CPU).")
     tbflags.add_loss_flags(parser)
     tbflags.add_optimizer_flags(parser)
     return parser
@@ -97,6 +101,35 @@ def _as_agent_output(outputs):
 
 
 INFERENCE_STATS = {"batches": 0, "env_steps": 0}
+
+
+def make_inference_runner(actor_model, inference_batcher, greedy=False):
+    """GIL-free C++ inference engine over the behavior model's parameters
+    (views of the flat actor buffer, so the learner's one-copy weight sync
+    covers the runner too). Shallow AtariNet only; the deep ResNet uses the
+    Python inference path."""
+    m = actor_model
+    weights = [
+        m.conv1.weight, m.conv1.bias,
+        m.conv2.weight, m.conv2.bias,
+        m.conv3.weight, m.conv3.bias,
+        m.fc.weight, m.fc.bias,
+        m.policy.weight, m.policy.bias,
+        m.baseline.weight, m.baseline.bias,
+    ]
+    num_layers = 0
+    if m.use_lstm:
+        num_layers = m.core.num_layers
+        for layer in range(num_layers):
+            weights += [
+                getattr(m.core, f"weight_ih_l{layer}"),
+                getattr(m.core, f"weight_hh_l{layer}"),
+                getattr(m.core, f"bias_ih_l{layer}"),
+                getattr(m.core, f"bias_hh_l{layer}"),
+            ]
+    return runtime._tbruntime.InferenceRunner(
+        inference_batcher, [w.detach() for w in weights], num_layers, greedy
+    )
 
 _inference_stream = None
 _inference_stream_lock = threading.Lock()
@@ -386,16 +419,28 @@ def train(flags):  # noqa: C901
         # Collectives must stay ordered: one learner thread under DP.
         for i in range(1 if world_size > 1 else flags.num_learner_threads)
     ]
-    inference_threads = [
-        threading.Thread(
-            target=inference,
-            name=f"inference-{i}",
-            args=(flags, inference_batcher, actor_model),
-        )
-        for i in range(flags.num_inference_threads)
-    ]
+    use_cpp_inference = (
+        flags.model == "shallow"
+        and flags.actor_device.type == "cuda"
+        and not getattr(flags, "py_inference", False)
+    )
+    inference_runner = None
+    if use_cpp_inference:
+        inference_runner = make_inference_runner(actor_model, inference_batcher)
+        inference_threads = []
+    else:
+        inference_threads = [
+            threading.Thread(
+                target=inference,
+                name=f"inference-{i}",
+                args=(flags, inference_batcher, actor_model),
+            )
+            for i in range(flags.num_inference_threads)
+        ]
 
     actorpool_thread.start()
+    if inference_runner is not None:
+        inference_runner.start(flags.num_inference_threads)
     for t in learner_threads + inference_threads:
         t.start()
 
@@ -452,6 +497,8 @@ def train(flags):  # noqa: C901
     finally:
         inference_batcher.close()
         learner_queue.close()
+        if inference_runner is not None:
+            inference_runner.stop()
         for t in learner_threads + inference_threads:
             t.join(timeout=10)
         actorpool_thread.join(timeout=10)

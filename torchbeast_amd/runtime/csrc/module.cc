@@ -14,6 +14,7 @@
 
 #include "actor_pool.h"
 #include "env_server.h"
+#include "inference_runner.h"
 #include "nest.h"
 #include "nest_pybind.h"
 #include "queues.h"
@@ -240,6 +241,20 @@ PYBIND11_MODULE(_tbruntime, m) {
            py::arg("initial_agent_state"))
       .def("run", &ActorPool::run, py::call_guard<py::gil_scoped_release>())
       .def("count", &ActorPool::count);
+
+  // ---- InferenceRunner ----
+  py::class_<InferenceRunner, std::shared_ptr<InferenceRunner>>(
+      m, "InferenceRunner")
+      .def(py::init<std::shared_ptr<DynamicBatcher>,
+                    std::vector<torch::Tensor>, int64_t, bool>(),
+           py::arg("inference_batcher"), py::arg("weights"),
+           py::arg("num_lstm_layers") = 0, py::arg("greedy") = false)
+      .def("start", &InferenceRunner::start, py::arg("num_threads") = 2,
+           py::call_guard<py::gil_scoped_release>())
+      .def("stop", &InferenceRunner::stop,
+           py::call_guard<py::gil_scoped_release>())
+      .def("batches", &InferenceRunner::batches)
+      .def("steps", &InferenceRunner::steps);
 
   // ---- EnvServer ----
   py::class_<EnvServer, std::shared_ptr<EnvServer>>(m, "Server")
