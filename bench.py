@@ -155,6 +155,7 @@ def main():
     learner_queue = runtime.BatchingQueue(
         batch_dim=1, minimum_batch_size=B, maximum_batch_size=B,
         maximum_queue_size=B,
+        output_device=str(device) if use_cuda else None,
     )
     inference_batcher = runtime.DynamicBatcher(
         batch_dim=1, minimum_batch_size=args.inference_min_batch_size,

@@ -325,6 +325,12 @@ def train(flags):  # noqa: C901
         maximum_batch_size=flags.batch_size,
         check_inputs=True,
         maximum_queue_size=flags.max_learner_queue_size,
+        # Batches are assembled straight into HBM (async DMA per rollout).
+        output_device=(
+            str(flags.learner_device)
+            if flags.learner_device.type == "cuda"
+            else None
+        ),
     )
     inference_batcher = runtime.DynamicBatcher(
         batch_dim=1,

@@ -90,9 +90,10 @@ class ActorPool {
         std::vector<const TensorNest*> steps;
         steps.reserve(rollout.size());
         for (const auto& s : rollout) steps.push_back(&s);
+        // Pinned slab so the learner-side dequeue can DMA it to the GPU.
         TensorNest stacked = TensorNest::apply_columns(
             steps, [](const std::vector<torch::Tensor>& column) {
-              return torch::cat(column, /*dim=*/0);
+              return cat_pinned(column, /*dim=*/0);
             });
         learner_queue_->enqueue(TensorNest(
             TensorNest::vector_t{std::move(stacked), rollout_initial_state}));

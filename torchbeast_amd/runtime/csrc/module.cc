@@ -162,13 +162,15 @@ PYBIND11_MODULE(_tbruntime, m) {
   // ---- BatchingQueue ----
   py::class_<BatchingQueue, std::shared_ptr<BatchingQueue>>(m, "BatchingQueue")
       .def(py::init<int64_t, std::optional<int64_t>, std::optional<int64_t>,
-                    std::optional<int64_t>, bool, std::optional<int64_t>>(),
+                    std::optional<int64_t>, bool, std::optional<int64_t>,
+                    std::optional<std::string>>(),
            py::arg("batch_dim") = 0,
            py::arg("minimum_batch_size") = std::nullopt,
            py::arg("maximum_batch_size") = std::nullopt,
            py::arg("timeout_ms") = std::nullopt,
            py::arg("check_inputs") = true,
-           py::arg("maximum_queue_size") = std::nullopt)
+           py::arg("maximum_queue_size") = std::nullopt,
+           py::arg("output_device") = std::nullopt)
       .def_property_readonly("batch_dim", &BatchingQueue::batch_dim)
       .def("enqueue", &BatchingQueue::enqueue, py::arg("nest"),
            py::call_guard<py::gil_scoped_release>())

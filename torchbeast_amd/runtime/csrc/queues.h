@@ -10,6 +10,8 @@
 
 #pragma once
 
+#include <ATen/cuda/CUDAContext.h>
+#include <ATen/cuda/CUDAEvent.h>
 #include <torch/extension.h>
 
 #include <atomic>
@@ -87,6 +89,34 @@ inline TensorNest batch_nests(const std::vector<const TensorNest*>& nests,
   return TensorNest::apply_columns(
       nests, [batch_dim](const std::vector<torch::Tensor>& column) {
         return cat_pinned(column, batch_dim);
+      });
+}
+
+// Batch assembly with the destination ON THE GPU: each (pinned) source
+// tensor is copied with one async H2D into its slice of a device-resident
+// output. Replaces {CPU cat -> one huge H2D} with B small DMAs that
+// interleave with inference traffic, and eliminates the host-side memcpy
+// entirely (the "rollouts DMA straight into the learner" design,
+// SURVEY.md §2.2).
+inline TensorNest batch_nests_to_device(
+    const std::vector<const TensorNest*>& nests, int64_t batch_dim,
+    torch::Device device) {
+  return TensorNest::apply_columns(
+      nests,
+      [batch_dim, device](const std::vector<torch::Tensor>& column) {
+        auto shape = column[0].sizes().vec();
+        int64_t total = 0;
+        for (const auto& t : column) total += t.size(batch_dim);
+        shape[batch_dim] = total;
+        torch::Tensor out =
+            torch::empty(shape, column[0].options().device(device));
+        int64_t offset = 0;
+        for (const auto& t : column) {
+          out.narrow(batch_dim, offset, t.size(batch_dim))
+              .copy_(t, /*non_blocking=*/true);
+          offset += t.size(batch_dim);
+        }
+        return out;
       });
 }
 
@@ -186,8 +216,13 @@ class BatchingQueue {
                 std::optional<int64_t> maximum_batch_size = std::nullopt,
                 std::optional<int64_t> timeout_ms = std::nullopt,
                 bool check_inputs = true,
-                std::optional<int64_t> maximum_queue_size = std::nullopt)
-      : batch_dim_(batch_dim),
+                std::optional<int64_t> maximum_queue_size = std::nullopt,
+                std::optional<std::string> output_device = std::nullopt)
+      : output_device_(output_device
+                           ? std::optional<torch::Device>(
+                                 torch::Device(*output_device))
+                           : std::nullopt),
+        batch_dim_(batch_dim),
         min_batch_size_(minimum_batch_size ? *minimum_batch_size : 1),
         max_batch_size_(maximum_batch_size
                             ? *maximum_batch_size
@@ -234,7 +269,27 @@ class BatchingQueue {
     std::vector<const TensorNest*> ptrs;
     ptrs.reserve(items.size());
     for (const auto& n : items) ptrs.push_back(&n);
-    auto result = std::make_pair(batch_nests(ptrs, batch_dim_),
+    TensorNest batched;
+    if (output_device_ && output_device_->is_cuda()) {
+      // Assemble directly on the GPU: async copies on a dedicated copy
+      // stream, then make the caller's stream wait on them.
+      if (!copy_stream_) {
+        copy_stream_ = std::make_unique<at::cuda::CUDAStream>(
+            at::cuda::getStreamFromPool(false, output_device_->index()));
+      }
+      at::cuda::CUDAStream current =
+          at::cuda::getCurrentCUDAStream(output_device_->index());
+      {
+        at::cuda::CUDAStreamGuard guard(*copy_stream_);
+        batched = batch_nests_to_device(ptrs, batch_dim_, *output_device_);
+      }
+      at::cuda::CUDAEvent ev;
+      ev.record(*copy_stream_);
+      ev.block(current);
+    } else {
+      batched = batch_nests(ptrs, batch_dim_);
+    }
+    auto result = std::make_pair(std::move(batched),
                                  static_cast<int64_t>(items.size()));
     wait_stats_.add(t1 - t0);
     cat_stats_.add(now_us() - t1);
@@ -256,12 +311,14 @@ class BatchingQueue {
   }
 
  private:
+  const std::optional<torch::Device> output_device_;
   const int64_t batch_dim_;
   const int64_t min_batch_size_;
   const int64_t max_batch_size_;
   const std::optional<std::chrono::milliseconds> timeout_;
   const bool check_inputs_;
   BoundedQueue<TensorNest> queue_;
+  std::unique_ptr<at::cuda::CUDAStream> copy_stream_;
   mutable StageStats wait_stats_;
   mutable StageStats cat_stats_;
 };
