@@ -104,30 +104,48 @@ class InferenceRunner {
     std::vector<torch::Tensor> state = top[1].flatten();  // [] or h,c [L,b,H]
 
     const int64_t b = frame.size(1);
+    // Quantize the compute batch to multiples of 64: MIOpen caches conv
+    // solutions per shape, and ragged dynamic-batch sizes would trigger a
+    // fresh find for every new size. Pad rows are zeroed and sliced off
+    // before sampling/D2H.
+    const int64_t bp = (b + 63) / 64 * 64;
     auto opts = torch::TensorOptions().device(device_);
 
-    torch::Tensor x = frame.to(opts.dtype(torch::kUInt8), /*non_blocking=*/true)
-                          .reshape({b, frame.size(2), frame.size(3),
-                                    frame.size(4)})
-                          .to(torch::kFloat32)
-                          .mul_(1.0f / 255.0f);
-    torch::Tensor rew = reward.to(opts.dtype(torch::kFloat32), true)
-                            .reshape({b, 1})
-                            .clamp_(-1, 1);
+    torch::Tensor frame_gpu =
+        frame.to(opts.dtype(torch::kUInt8), /*non_blocking=*/true)
+            .reshape({b, frame.size(2), frame.size(3), frame.size(4)});
+    torch::Tensor x = torch::empty(
+        {bp, frame.size(2), frame.size(3), frame.size(4)},
+        opts.dtype(torch::kFloat32));
+    x.narrow(0, 0, b).copy_(frame_gpu).mul_(1.0f / 255.0f);
+    if (bp > b) x.narrow(0, b, bp - b).zero_();
+
+    torch::Tensor rew = torch::zeros({bp, 1}, opts.dtype(torch::kFloat32));
+    rew.narrow(0, 0, b)
+        .copy_(reward.to(opts.dtype(torch::kFloat32), true).reshape({b, 1}))
+        .clamp_(-1, 1);
 
     x = at::conv2d(x, weights_[0], weights_[1], /*stride=*/4).relu_();
     x = at::conv2d(x, weights_[2], weights_[3], 2).relu_();
     x = at::conv2d(x, weights_[4], weights_[5], 1).relu_();
-    x = at::linear(x.reshape({b, -1}), weights_[6], weights_[7]).relu_();
+    x = at::linear(x.reshape({bp, -1}), weights_[6], weights_[7]).relu_();
     torch::Tensor core = at::cat({x, rew}, 1);
 
     std::vector<torch::Tensor> new_state_gpu;
     if (num_lstm_layers_ > 0) {
       TORCH_CHECK(state.size() == 2, "lstm runner needs (h, c) state");
+      const int64_t L = state[0].size(0);
+      const int64_t H = state[0].size(2);
       torch::Tensor nd = (~done.reshape({1, b, 1}))
                              .to(opts.dtype(torch::kFloat32), true);
-      torch::Tensor h = state[0].to(opts.dtype(torch::kFloat32), true) * nd;
-      torch::Tensor c = state[1].to(opts.dtype(torch::kFloat32), true) * nd;
+      auto pad_state = [&](const torch::Tensor& s) {
+        torch::Tensor g = torch::zeros({L, bp, H}, opts.dtype(torch::kFloat32));
+        g.narrow(1, 0, b).copy_(
+            s.to(opts.dtype(torch::kFloat32), true) * nd);
+        return g;
+      };
+      torch::Tensor h = pad_state(state[0]);
+      torch::Tensor c = pad_state(state[1]);
       torch::Tensor layer_in = core;
       std::vector<torch::Tensor> hs, cs;
       for (int64_t l = 0; l < num_lstm_layers_; ++l) {
@@ -149,17 +167,23 @@ class InferenceRunner {
         layer_in = h_new;
       }
       core = layer_in;
-      new_state_gpu = {at::stack(hs), at::stack(cs)};
+      new_state_gpu = {at::stack(hs).narrow(1, 0, b),
+                       at::stack(cs).narrow(1, 0, b)};
     }
 
-    torch::Tensor logits = at::linear(core, weights_[8], weights_[9]);
+    torch::Tensor logits =
+        at::linear(core, weights_[8], weights_[9]).narrow(0, 0, b);
     torch::Tensor baseline =
-        at::linear(core, weights_[10], weights_[11]).reshape({b});
+        at::linear(core, weights_[10], weights_[11]).narrow(0, 0, b).reshape(
+            {b});
     torch::Tensor action;
     if (greedy_) {
       action = at::argmax(logits, -1);
     } else {
-      action = at::multinomial(at::softmax(logits, -1), 1).reshape({b});
+      // Gumbel-argmax categorical sampling: exactly softmax(logits), no
+      // host sync, three elementwise kernels (at::multinomial stalls).
+      torch::Tensor u = at::rand_like(logits).clamp_(1e-20, 1.0);
+      action = at::argmax(logits - at::log(-at::log(u)), -1);
     }
 
     // D2H into pinned host tensors, one stream sync for the whole batch.
