@@ -23,7 +23,10 @@ ext_modules = [
     # ops from C++ (GIL-free inference engine).
     CUDAExtension(
         name="torchbeast_amd.runtime._tbruntime",
-        sources=["torchbeast_amd/runtime/csrc/module.cc"],
+        sources=[
+            "torchbeast_amd/runtime/csrc/module.cc",
+            "torchbeast_amd/ops/hip/atari_trunk.hip",
+        ],
         extra_compile_args={
             "cxx": ["-O3", "-std=c++17", "-pthread"],
             "nvcc": ["-O3", "-std=c++17"],

@@ -177,6 +177,47 @@ def test_lstm_input_grads_match():
     torch.testing.assert_close(x_g.grad.cpu(), x_c.grad, rtol=1e-4, atol=1e-4)
 
 
+@pytest.mark.parametrize("N,shape", [(7, (4, 84, 84)), (3, (4, 36, 36)),
+                                     (64, (4, 84, 84))])
+def test_atari_trunk_matches_eager(N, shape):
+    import torch.nn.functional as Fn
+
+    torch.manual_seed(6)
+    conv1 = torch.nn.Conv2d(shape[0], 32, 8, stride=4).cuda()
+    conv2 = torch.nn.Conv2d(32, 64, 4, stride=2).cuda()
+    conv3 = torch.nn.Conv2d(64, 64, 3, stride=1).cuda()
+    frames = torch.randint(0, 256, (N, *shape), dtype=torch.uint8).cuda()
+
+    out = tbops.atari_trunk(frames, conv1, conv2, conv3)
+    assert out is not None
+
+    x = frames.float() / 255.0
+    ref = Fn.relu(conv3(Fn.relu(conv2(Fn.relu(conv1(x)))))).view(N, -1)
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-4)
+
+    # Weight/bias gradients vs autograd through the eager chain.
+    out.square().sum().backward()
+    fused_grads = [p.grad.clone() for c in (conv1, conv2, conv3)
+                   for p in (c.weight, c.bias)]
+    for c in (conv1, conv2, conv3):
+        c.weight.grad = None
+        c.bias.grad = None
+    ref2 = Fn.relu(conv3(Fn.relu(conv2(Fn.relu(conv1(x)))))).view(N, -1)
+    ref2.square().sum().backward()
+    eager_grads = [p.grad for c in (conv1, conv2, conv3)
+                   for p in (c.weight, c.bias)]
+    for fg, eg in zip(fused_grads, eager_grads):
+        torch.testing.assert_close(fg, eg, rtol=2e-3, atol=1e-3)
+
+
+def test_atari_trunk_rejects_oversized_frames():
+    import torchbeast_amd.ops as ops
+
+    ext = ops.require_ext()
+    assert ext.atari_trunk_supported(4, 84, 84)
+    assert not ext.atari_trunk_supported(3, 210, 160)
+
+
 def test_model_forward_gpu_runs_with_fused_ops():
     from torchbeast_amd.models import AtariNet
 

@@ -62,13 +62,20 @@ class AtariNet(nn.Module):
         )
 
     def forward(self, inputs, core_state=()):
+        from torchbeast_amd.ops import functional as tbf
+
         frame = inputs["frame"]  # [T, B, C, H, W], uint8.
         T, B = frame.shape[:2]
-        x = torch.flatten(frame, 0, 1).float() / 255.0
-        x = F.relu(self.conv1(x))
-        x = F.relu(self.conv2(x))
-        x = F.relu(self.conv3(x))
-        x = F.relu(self.fc(x.view(T * B, -1)))
+        flat_frames = torch.flatten(frame, 0, 1)
+        # GPU: fused u8 conv trunk (one kernel); CPU/oversized: eager chain.
+        x = tbf.atari_trunk(flat_frames, self.conv1, self.conv2, self.conv3)
+        if x is None:
+            x = flat_frames.float() / 255.0
+            x = F.relu(self.conv1(x))
+            x = F.relu(self.conv2(x))
+            x = F.relu(self.conv3(x))
+            x = x.view(T * B, -1)
+        x = F.relu(self.fc(x))
 
         clipped_reward = torch.clamp(inputs["reward"], -1, 1).view(T * B, 1)
         if self.use_last_action:

@@ -256,6 +256,51 @@ def rmsprop_step(param, grad, square_avg, lr, alpha, eps, clip_norm=None):
 
 
 # ---------------------------------------------------------------------------
+# Fused AtariNet conv trunk (u8 frames -> flat conv features).
+# ---------------------------------------------------------------------------
+
+
+class _AtariTrunk(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, frames, w1, b1, w2, b2, w3, b3):
+        ext = _ext_for(frames, "atari_trunk_fwd")
+        out3, save1, save2 = ext.atari_trunk_fwd(
+            frames, w1, b1, w2, b2, w3, b3, True
+        )
+        ctx.save_for_backward(frames, w1, w2, w3, save1, save2, out3)
+        return out3
+
+    @staticmethod
+    def backward(ctx, d_out3):
+        frames, w1, w2, w3, save1, save2, out3 = ctx.saved_tensors
+        ext = ops_mod.require_ext()
+        dw1, db1, dw2, db2, dw3, db3 = ext.atari_trunk_bwd(
+            frames, w1, w2, w3, save1, save2, out3, d_out3
+        )
+        return None, dw1, db1, dw2, db2, dw3, db3
+
+
+def atari_trunk(frames, conv1, conv2, conv3):
+    """Fused u8-frame conv trunk for GPU [N,C,H,W] u8 frames.
+
+    Returns flat post-relu conv3 features [N, 64*H3*W3], or None when the
+    fused kernel doesn't apply (CPU tensors, or geometry exceeding LDS,
+    e.g. 210x160 full-res frames) — caller falls back to eager convs.
+    """
+    if not frames.is_cuda or frames.dtype != torch.uint8:
+        return None
+    ext = _ext_for(frames, "atari_trunk_fwd")
+    if ext is None or not ext.atari_trunk_supported(*frames.shape[1:]):
+        return None
+    args = (frames.contiguous(), conv1.weight, conv1.bias, conv2.weight,
+            conv2.bias, conv3.weight, conv3.bias)
+    if torch.is_grad_enabled():
+        return _AtariTrunk.apply(*args)
+    (out3,) = ext.atari_trunk_fwd(*args, False)
+    return out3
+
+
+# ---------------------------------------------------------------------------
 # Policy sampling: softmax + multinomial (train) / argmax (eval).
 # ---------------------------------------------------------------------------
 

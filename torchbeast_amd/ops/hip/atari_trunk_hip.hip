@@ -1,0 +1,394 @@
+// Fused AtariNet conv trunk — hand-written CDNA4 (gfx950) kernels.
+//
+// Replaces the reference's three stock conv2d calls + frame normalize
+// (ref: torchbeast/monobeast.py:552-559, 586-589), which on ROCm expand to
+// ~12 MIOpen/im2col/transpose launches per forward. Design:
+//
+// - ONE workgroup per sample; the whole conv chain runs inside the CU with
+//   every intermediate activation in LDS (84x84x4 frame: u8 input 28 KB +
+//   conv1 out 50 KB + conv2 out 20 KB < 160 KB LDS). Weights (w1 32 KB,
+//   w2 128 KB, w3 144 KB) stay L2/L3-resident and are shared by all
+//   concurrent workgroups.
+// - u8 -> f32 /255 is folded into the conv1 tap sum (integer frame bytes
+//   scaled once per output), so the [N,C,H,W] f32 frame tensor the
+//   reference materializes never exists.
+// - Backward: a per-sample dgrad kernel walks the chain in reverse with
+//   relu masks from the saved activations, writing masked per-layer grads;
+//   weight gradients are block-per-(co,ky,kx) reduction kernels over
+//   [N, out-positions] (no atomics: one block owns one weight slice).
+//
+// Shapes are runtime parameters; geometry checks live in
+// atari_trunk_supported(). 256-thread blocks (4 wavefronts).
+
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include "atari_trunk.h"
+
+namespace tbamd {
+
+namespace {
+
+constexpr int kThreads = 256;
+constexpr int64_t kLdsBudget = 160 * 1024;
+
+struct Geom {
+  int C, H, W;
+  int H1, W1;  // conv1 out (k8 s4), 32 ch
+  int H2, W2;  // conv2 out (k4 s2), 64 ch
+  int H3, W3;  // conv3 out (k3 s1), 64 ch
+  size_t lds_in, lds1, lds2, lds_total;
+};
+
+Geom make_geom(int64_t C, int64_t H, int64_t W) {
+  Geom g;
+  g.C = C; g.H = H; g.W = W;
+  g.H1 = (H - 8) / 4 + 1; g.W1 = (W - 8) / 4 + 1;
+  g.H2 = (g.H1 - 4) / 2 + 1; g.W2 = (g.W1 - 4) / 2 + 1;
+  g.H3 = g.H2 - 2; g.W3 = g.W2 - 2;
+  g.lds_in = ((size_t)C * H * W + 15) & ~size_t(15);
+  g.lds1 = (size_t)32 * g.H1 * g.W1 * 4;
+  g.lds2 = (size_t)64 * g.H2 * g.W2 * 4;
+  g.lds_total = g.lds_in + g.lds1 + g.lds2;
+  return g;
+}
+
+__global__ __launch_bounds__(kThreads) void trunk_fwd_kernel(
+    const uint8_t* __restrict__ frames, const float* __restrict__ w1,
+    const float* __restrict__ b1, const float* __restrict__ w2,
+    const float* __restrict__ b2, const float* __restrict__ w3,
+    const float* __restrict__ b3, int C, int H, int W, int H1, int W1,
+    int H2, int W2, int H3, int W3, size_t lds_in_bytes,
+    float* __restrict__ out3, float* __restrict__ save1,
+    float* __restrict__ save2) {
+  extern __shared__ unsigned char smem[];
+  uint8_t* s_in = smem;
+  float* s_out1 = reinterpret_cast<float*>(smem + lds_in_bytes);
+  float* s_out2 = s_out1 + 32 * H1 * W1;
+
+  const int n = blockIdx.x;
+  const int tid = threadIdx.x;
+
+  {  // Stage the u8 frame into LDS (word-wide where possible).
+    const uint8_t* f = frames + (int64_t)n * C * H * W;
+    const int total = C * H * W;
+    const int words = total / 4;
+    const uint32_t* f32p = reinterpret_cast<const uint32_t*>(f);
+    uint32_t* s32p = reinterpret_cast<uint32_t*>(s_in);
+    for (int i = tid; i < words; i += kThreads) s32p[i] = f32p[i];
+    for (int i = words * 4 + tid; i < total; i += kThreads) s_in[i] = f[i];
+  }
+  __syncthreads();
+
+  {  // conv1: k8 s4, C -> 32, fused /255 normalize.
+    const int n1 = 32 * H1 * W1;
+    const int plane1 = H1 * W1;
+    for (int i = tid; i < n1; i += kThreads) {
+      const int co = i / plane1;
+      const int r = i - co * plane1;
+      const int oy = r / W1, ox = r - (r / W1) * W1;
+      float acc = 0.f;
+      const float* wco = w1 + (int64_t)co * C * 64;
+      for (int ci = 0; ci < C; ++ci) {
+        const uint8_t* in_c = s_in + ci * H * W + (oy * 4) * W + ox * 4;
+        const float* w_c = wco + ci * 64;
+#pragma unroll
+        for (int ky = 0; ky < 8; ++ky) {
+#pragma unroll
+          for (int kx = 0; kx < 8; ++kx) {
+            acc += (float)in_c[ky * W + kx] * w_c[ky * 8 + kx];
+          }
+        }
+      }
+      const float v = fmaxf(acc * (1.f / 255.f) + b1[co], 0.f);
+      s_out1[i] = v;
+      if (save1 != nullptr) save1[(int64_t)n * n1 + i] = v;
+    }
+  }
+  __syncthreads();
+
+  {  // conv2: k4 s2, 32 -> 64.
+    const int n2 = 64 * H2 * W2;
+    const int plane2 = H2 * W2;
+    for (int i = tid; i < n2; i += kThreads) {
+      const int co = i / plane2;
+      const int r = i - co * plane2;
+      const int oy = r / W2, ox = r - (r / W2) * W2;
+      float acc = b2[co];
+      const float* wco = w2 + (int64_t)co * 32 * 16;
+      const float* in_base = s_out1 + (oy * 2) * W1 + ox * 2;
+      for (int ci = 0; ci < 32; ++ci) {
+        const float* in_c = in_base + ci * H1 * W1;
+        const float* w_c = wco + ci * 16;
+#pragma unroll
+        for (int ky = 0; ky < 4; ++ky) {
+#pragma unroll
+          for (int kx = 0; kx < 4; ++kx) {
+            acc += in_c[ky * W1 + kx] * w_c[ky * 4 + kx];
+          }
+        }
+      }
+      const float v = fmaxf(acc, 0.f);
+      s_out2[i] = v;
+      if (save2 != nullptr) save2[(int64_t)n * n2 + i] = v;
+    }
+  }
+  __syncthreads();
+
+  {  // conv3: k3 s1, 64 -> 64, straight to global (it's the fc input).
+    const int n3 = 64 * H3 * W3;
+    const int plane3 = H3 * W3;
+    for (int i = tid; i < n3; i += kThreads) {
+      const int co = i / plane3;
+      const int r = i - co * plane3;
+      const int oy = r / W3, ox = r - (r / W3) * W3;
+      float acc = b3[co];
+      const float* wco = w3 + (int64_t)co * 64 * 9;
+      const float* in_base = s_out2 + oy * W2 + ox;
+      for (int ci = 0; ci < 64; ++ci) {
+        const float* in_c = in_base + ci * H2 * W2;
+        const float* w_c = wco + ci * 9;
+#pragma unroll
+        for (int ky = 0; ky < 3; ++ky) {
+#pragma unroll
+          for (int kx = 0; kx < 3; ++kx) {
+            acc += in_c[ky * W2 + kx] * w_c[ky * 3 + kx];
+          }
+        }
+      }
+      out3[(int64_t)n * n3 + i] = fmaxf(acc, 0.f);
+    }
+  }
+}
+
+// Reverse chain per sample: d3 = d_out3 * relu'(out3); d2 = conv3^T(d3)
+// * relu'(out2); d1 = conv2^T(d2) * relu'(out1). d3/d2/d1 are written to
+// global for the wgrad kernels. LDS: d3 + d2 planes.
+__global__ __launch_bounds__(kThreads) void trunk_bwd_dgrad_kernel(
+    const float* __restrict__ d_out3, const float* __restrict__ out3,
+    const float* __restrict__ out2, const float* __restrict__ out1,
+    const float* __restrict__ w3, const float* __restrict__ w2, int H1,
+    int W1, int H2, int W2, int H3, int W3, float* __restrict__ g3,
+    float* __restrict__ g2, float* __restrict__ g1) {
+  extern __shared__ unsigned char smem[];
+  float* s_d3 = reinterpret_cast<float*>(smem);          // 64*H3*W3
+  float* s_d2 = s_d3 + 64 * H3 * W3;                     // 64*H2*W2
+
+  const int n = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int n3 = 64 * H3 * W3;
+  const int n2 = 64 * H2 * W2;
+  const int n1 = 32 * H1 * W1;
+
+  for (int i = tid; i < n3; i += kThreads) {
+    const float v = out3[(int64_t)n * n3 + i] > 0.f
+                        ? d_out3[(int64_t)n * n3 + i]
+                        : 0.f;
+    s_d3[i] = v;
+    g3[(int64_t)n * n3 + i] = v;
+  }
+  __syncthreads();
+
+  {  // conv3 dgrad (k3 s1): d2[ci,y,x] = sum_{co,ky,kx} d3[co,y-ky,x-kx]*w3.
+    const int plane2 = H2 * W2;
+    for (int i = tid; i < n2; i += kThreads) {
+      const int ci = i / plane2;
+      const int r = i - ci * plane2;
+      const int y = r / W2, x = r - (r / W2) * W2;
+      float acc = 0.f;
+      for (int co = 0; co < 64; ++co) {
+        const float* d3_c = s_d3 + co * H3 * W3;
+        const float* w_c = w3 + ((int64_t)co * 64 + ci) * 9;
+#pragma unroll
+        for (int ky = 0; ky < 3; ++ky) {
+          const int oy = y - ky;
+          if (oy < 0 || oy >= H3) continue;
+#pragma unroll
+          for (int kx = 0; kx < 3; ++kx) {
+            const int ox = x - kx;
+            if (ox < 0 || ox >= W3) continue;
+            acc += d3_c[oy * W3 + ox] * w_c[ky * 3 + kx];
+          }
+        }
+      }
+      const float v = out2[(int64_t)n * n2 + i] > 0.f ? acc : 0.f;
+      s_d2[i] = v;
+      g2[(int64_t)n * n2 + i] = v;
+    }
+  }
+  __syncthreads();
+
+  {  // conv2 dgrad (k4 s2): d1[ci,y,x] over taps with (y-ky)%2==0.
+    const int plane1 = H1 * W1;
+    for (int i = tid; i < n1; i += kThreads) {
+      const int ci = i / plane1;
+      const int r = i - ci * plane1;
+      const int y = r / W1, x = r - (r / W1) * W1;
+      float acc = 0.f;
+      for (int co = 0; co < 64; ++co) {
+        const float* d2_c = s_d2 + co * H2 * W2;
+        const float* w_c = w2 + ((int64_t)co * 32 + ci) * 16;
+#pragma unroll
+        for (int ky = 0; ky < 4; ++ky) {
+          const int ty = y - ky;
+          if (ty < 0 || (ty & 1) || ty / 2 >= H2) continue;
+#pragma unroll
+          for (int kx = 0; kx < 4; ++kx) {
+            const int tx = x - kx;
+            if (tx < 0 || (tx & 1) || tx / 2 >= W2) continue;
+            acc += d2_c[(ty / 2) * W2 + tx / 2] * w_c[ky * 4 + kx];
+          }
+        }
+      }
+      const float v = out1[(int64_t)n * n1 + i] > 0.f ? acc : 0.f;
+      g1[(int64_t)n * n1 + i] = v;
+    }
+  }
+}
+
+// dw[co,ci,ky,kx] = sum_n sum_{oy,ox} d[n,co,oy,ox] * x[n,ci,oy*s+ky,ox*s+kx]
+// One block per (co,ky,kx); threads split as (ci, n-group); LDS-reduced.
+// IN_T is float (activations) or uint8_t (frames, with 1/255 fold).
+template <typename IN_T>
+__global__ __launch_bounds__(kThreads) void conv_wgrad_kernel(
+    const float* __restrict__ d, const IN_T* __restrict__ x, int N, int Cin,
+    int Hx, int Wx, int Ho, int Wo, int K, int stride, float scale,
+    float* __restrict__ dw) {
+  __shared__ float red[kThreads];
+  const int b = blockIdx.x;  // co*K*K + ky*K + kx
+  const int co = b / (K * K);
+  const int ky = (b / K) % K;
+  const int kx = b % K;
+
+  const int groups = kThreads / Cin;  // Cin in {4, 32, 64} divides 256.
+  const int ci = threadIdx.x % Cin;
+  const int grp = threadIdx.x / Cin;
+
+  float acc = 0.f;
+  if (grp < groups) {
+    const int64_t Cout_plane = (int64_t)Ho * Wo;
+    for (int n = grp; n < N; n += groups) {
+      const float* d_nc = d + ((int64_t)n * gridDim.x / (K * K) + co) * Cout_plane;
+      const IN_T* x_nc = x + ((int64_t)n * Cin + ci) * Hx * Wx + ky * Wx + kx;
+      for (int oy = 0; oy < Ho; ++oy) {
+        const float* d_row = d_nc + oy * Wo;
+        const IN_T* x_row = x_nc + (oy * stride) * Wx;
+        for (int ox = 0; ox < Wo; ++ox) {
+          acc += d_row[ox] * (float)x_row[ox * stride];
+        }
+      }
+    }
+  }
+  red[threadIdx.x] = acc;
+  __syncthreads();
+  // Reduce over n-groups for each ci: entries ci, ci+Cin, ci+2*Cin, ...
+  if (threadIdx.x < Cin) {
+    float sum = 0.f;
+    for (int g = 0; g < groups; ++g) sum += red[g * Cin + threadIdx.x];
+    dw[((int64_t)co * Cin + threadIdx.x) * K * K + ky * K + kx] = sum * scale;
+  }
+}
+
+}  // namespace
+
+bool atari_trunk_supported(int64_t C, int64_t H, int64_t W) {
+  if (H < 36 || W < 36 || C < 1 || C > 16) return false;
+  Geom g = make_geom(C, H, W);
+  if (g.H3 < 1 || g.W3 < 1) return false;
+  size_t bwd_lds = ((size_t)64 * g.H3 * g.W3 + (size_t)64 * g.H2 * g.W2) * 4;
+  return g.lds_total <= kLdsBudget && bwd_lds <= kLdsBudget;
+}
+
+std::vector<torch::Tensor> atari_trunk_fwd(
+    torch::Tensor frames, torch::Tensor w1, torch::Tensor b1,
+    torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
+    bool save_for_backward) {
+  TORCH_CHECK(frames.is_cuda() && frames.scalar_type() == torch::kUInt8,
+              "atari_trunk_fwd: frames must be u8 on GPU");
+  TORCH_CHECK(frames.dim() == 4, "frames must be [N,C,H,W]");
+  auto fr = frames.contiguous();
+  const int64_t N = fr.size(0);
+  Geom g = make_geom(fr.size(1), fr.size(2), fr.size(3));
+  TORCH_CHECK(atari_trunk_supported(fr.size(1), fr.size(2), fr.size(3)),
+              "frame geometry unsupported by fused trunk");
+
+  auto opts = w1.options();
+  auto out3 = torch::empty({N, (int64_t)64 * g.H3 * g.W3}, opts);
+  torch::Tensor save1, save2;
+  float* save1_p = nullptr;
+  float* save2_p = nullptr;
+  if (save_for_backward) {
+    save1 = torch::empty({N, 32, g.H1, g.W1}, opts);
+    save2 = torch::empty({N, 64, g.H2, g.W2}, opts);
+    save1_p = save1.data_ptr<float>();
+    save2_p = save2.data_ptr<float>();
+  }
+  if (N == 0) return save_for_backward
+                  ? std::vector<torch::Tensor>{out3, save1, save2}
+                  : std::vector<torch::Tensor>{out3};
+
+  auto w1c = w1.contiguous(); auto b1c = b1.contiguous();
+  auto w2c = w2.contiguous(); auto b2c = b2.contiguous();
+  auto w3c = w3.contiguous(); auto b3c = b3.contiguous();
+  hipLaunchKernelGGL(trunk_fwd_kernel, dim3(N), dim3(kThreads), g.lds_total,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     fr.data_ptr<uint8_t>(), w1c.data_ptr<float>(),
+                     b1c.data_ptr<float>(), w2c.data_ptr<float>(),
+                     b2c.data_ptr<float>(), w3c.data_ptr<float>(),
+                     b3c.data_ptr<float>(), g.C, g.H, g.W, g.H1, g.W1, g.H2,
+                     g.W2, g.H3, g.W3, g.lds_in, out3.data_ptr<float>(),
+                     save1_p, save2_p);
+  if (save_for_backward) return {out3, save1, save2};
+  return {out3};
+}
+
+std::vector<torch::Tensor> atari_trunk_bwd(
+    torch::Tensor frames, torch::Tensor w1, torch::Tensor w2,
+    torch::Tensor w3, torch::Tensor out1, torch::Tensor out2,
+    torch::Tensor out3_flat, torch::Tensor d_out3) {
+  auto fr = frames.contiguous();
+  const int64_t N = fr.size(0);
+  Geom g = make_geom(fr.size(1), fr.size(2), fr.size(3));
+  auto opts = w1.options();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+
+  auto g3 = torch::empty({N, 64, g.H3, g.W3}, opts);
+  auto g2 = torch::empty({N, 64, g.H2, g.W2}, opts);
+  auto g1 = torch::empty({N, 32, g.H1, g.W1}, opts);
+
+  const size_t lds = ((size_t)64 * g.H3 * g.W3 + (size_t)64 * g.H2 * g.W2) * 4;
+  hipLaunchKernelGGL(trunk_bwd_dgrad_kernel, dim3(N), dim3(kThreads), lds,
+                     stream, d_out3.contiguous().data_ptr<float>(),
+                     out3_flat.data_ptr<float>(), out2.data_ptr<float>(),
+                     out1.data_ptr<float>(), w3.contiguous().data_ptr<float>(),
+                     w2.contiguous().data_ptr<float>(), g.H1, g.W1, g.H2,
+                     g.W2, g.H3, g.W3, g3.data_ptr<float>(),
+                     g2.data_ptr<float>(), g1.data_ptr<float>());
+
+  auto dw3 = torch::empty_like(w3);
+  auto dw2 = torch::empty_like(w2);
+  auto dw1 = torch::empty_like(w1);
+  // conv3 wgrad: d=g3, x=out2.
+  hipLaunchKernelGGL(conv_wgrad_kernel<float>, dim3(64 * 9), dim3(kThreads),
+                     0, stream, g3.data_ptr<float>(), out2.data_ptr<float>(),
+                     N, 64, g.H2, g.W2, g.H3, g.W3, 3, 1, 1.f,
+                     dw3.data_ptr<float>());
+  // conv2 wgrad: d=g2, x=out1.
+  hipLaunchKernelGGL(conv_wgrad_kernel<float>, dim3(64 * 16), dim3(kThreads),
+                     0, stream, g2.data_ptr<float>(), out1.data_ptr<float>(),
+                     N, 32, g.H1, g.W1, g.H2, g.W2, 4, 2, 1.f,
+                     dw2.data_ptr<float>());
+  // conv1 wgrad: d=g1, x=frames u8 (fold the 1/255 normalize).
+  hipLaunchKernelGGL(conv_wgrad_kernel<uint8_t>, dim3(32 * 64), dim3(kThreads),
+                     0, stream, g1.data_ptr<float>(), fr.data_ptr<uint8_t>(),
+                     N, g.C, g.H, g.W, g.H1, g.W1, 8, 4, 1.f / 255.f,
+                     dw1.data_ptr<float>());
+
+  auto db3 = g3.sum({0, 2, 3});
+  auto db2 = g2.sum({0, 2, 3});
+  auto db1 = g1.sum({0, 2, 3});
+  return {dw1, db1, dw2, db2, dw3, db3};
+}
+
+}  // namespace tbamd

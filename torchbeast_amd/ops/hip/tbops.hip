@@ -25,6 +25,8 @@
 
 #include <vector>
 
+#include "atari_trunk.h"
+
 namespace cg = cooperative_groups;
 
 #define DEVCHECK(x) TORCH_CHECK(x == hipSuccess, "HIP error: ", hipGetErrorString(x))
@@ -715,6 +717,9 @@ std::vector<torch::Tensor> lstm_unroll_bwd(
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("atari_trunk_fwd", &tbamd::atari_trunk_fwd);
+  m.def("atari_trunk_bwd", &tbamd::atari_trunk_bwd);
+  m.def("atari_trunk_supported", &tbamd::atari_trunk_supported);
   m.def("vtrace_from_logits", &vtrace_from_logits);
   m.def("fused_impala_loss_fwd", &fused_impala_loss_fwd);
   m.def("rmsprop_step", &rmsprop_step);

@@ -20,6 +20,7 @@
 #include <thread>
 #include <vector>
 
+#include "../../ops/hip/atari_trunk.h"
 #include "queues.h"
 
 namespace tbruntime {
@@ -111,24 +112,32 @@ class InferenceRunner {
     const int64_t bp = (b + 63) / 64 * 64;
     auto opts = torch::TensorOptions().device(device_);
 
-    torch::Tensor frame_gpu =
-        frame.to(opts.dtype(torch::kUInt8), /*non_blocking=*/true)
-            .reshape({b, frame.size(2), frame.size(3), frame.size(4)});
-    torch::Tensor x = torch::empty(
-        {bp, frame.size(2), frame.size(3), frame.size(4)},
-        opts.dtype(torch::kFloat32));
-    x.narrow(0, 0, b).copy_(frame_gpu).mul_(1.0f / 255.0f);
-    if (bp > b) x.narrow(0, b, bp - b).zero_();
+    const int64_t C = frame.size(2), H = frame.size(3), W = frame.size(4);
+    torch::Tensor frames_p = torch::zeros({bp, C, H, W},
+                                          opts.dtype(torch::kUInt8));
+    frames_p.narrow(0, 0, b).copy_(frame.reshape({b, C, H, W}),
+                                   /*non_blocking=*/true);
 
     torch::Tensor rew = torch::zeros({bp, 1}, opts.dtype(torch::kFloat32));
     rew.narrow(0, 0, b)
         .copy_(reward.to(opts.dtype(torch::kFloat32), true).reshape({b, 1}))
         .clamp_(-1, 1);
 
-    x = at::conv2d(x, weights_[0], weights_[1], /*stride=*/4).relu_();
-    x = at::conv2d(x, weights_[2], weights_[3], 2).relu_();
-    x = at::conv2d(x, weights_[4], weights_[5], 1).relu_();
-    x = at::linear(x.reshape({bp, -1}), weights_[6], weights_[7]).relu_();
+    torch::Tensor x;
+    if (tbamd::atari_trunk_supported(C, H, W)) {
+      // Hand-written fused CDNA4 conv trunk: one kernel for the u8
+      // normalize + 3 convs.
+      x = tbamd::atari_trunk_fwd(frames_p, weights_[0], weights_[1],
+                                 weights_[2], weights_[3], weights_[4],
+                                 weights_[5], /*save_for_backward=*/false)[0];
+    } else {
+      x = frames_p.to(torch::kFloat32).mul_(1.0f / 255.0f);
+      x = at::conv2d(x, weights_[0], weights_[1], /*stride=*/4).relu_();
+      x = at::conv2d(x, weights_[2], weights_[3], 2).relu_();
+      x = at::conv2d(x, weights_[4], weights_[5], 1).relu_();
+      x = x.reshape({bp, -1});
+    }
+    x = at::linear(x, weights_[6], weights_[7]).relu_();
     torch::Tensor core = at::cat({x, rew}, 1);
 
     std::vector<torch::Tensor> new_state_gpu;
