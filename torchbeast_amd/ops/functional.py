@@ -272,11 +272,28 @@ class _AtariTrunk(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, d_out3):
+        # Backward = relu masks + library convolution_backward per layer
+        # (GEMM-shaped reductions; see ops/hip/atari_trunk.hip for the
+        # hand-written variant kept for reference/benchmarks).
         frames, w1, w2, w3, save1, save2, out3 = ctx.saved_tensors
-        ext = ops_mod.require_ext()
-        dw1, db1, dw2, db2, dw3, db3 = ext.atari_trunk_bwd(
-            frames, w1, w2, w3, save1, save2, out3, d_out3
-        )
+        N = frames.shape[0]
+        H2, W2 = save2.shape[2], save2.shape[3]
+        d3 = (d_out3 * (out3 > 0)).view(N, 64, H2 - 2, W2 - 2)
+
+        def conv_bwd(grad_out, inp, weight, stride, need_input_grad):
+            return torch.ops.aten.convolution_backward(
+                grad_out, inp, weight,
+                [weight.shape[0]],  # bias sizes
+                [stride, stride], [0, 0], [1, 1], False, [0, 0], 1,
+                [need_input_grad, True, True],
+            )
+
+        d2_pre, dw3, db3 = conv_bwd(d3, save2, w3, 1, True)
+        d2 = d2_pre * (save2 > 0)
+        d1_pre, dw2, db2 = conv_bwd(d2, save1, w2, 2, True)
+        d1 = d1_pre * (save1 > 0)
+        frames_f = frames.float().mul_(1.0 / 255.0)
+        _, dw1, db1 = conv_bwd(d1, frames_f, w1, 4, False)
         return None, dw1, db1, dw2, db2, dw3, db3
 
 

@@ -30,7 +30,8 @@ namespace tbamd {
 
 namespace {
 
-constexpr int kThreads = 256;
+constexpr int kThreads = 256;      // wgrad/dgrad kernels
+constexpr int kFwdThreads = 512;   // fwd kernel: 8 waves/CU for latency hiding
 constexpr int64_t kLdsBudget = 160 * 1024;
 
 struct Geom {
@@ -50,11 +51,12 @@ Geom make_geom(int64_t C, int64_t H, int64_t W) {
   g.lds_in = ((size_t)C * H * W + 15) & ~size_t(15);
   g.lds1 = (size_t)32 * g.H1 * g.W1 * 4;
   g.lds2 = (size_t)64 * g.H2 * g.W2 * 4;
-  g.lds_total = g.lds_in + g.lds1 + g.lds2;
+  // + w1 staged in LDS (32 x C x 8 x 8 fp32).
+  g.lds_total = g.lds_in + g.lds1 + g.lds2 + (size_t)32 * C * 64 * 4;
   return g;
 }
 
-__global__ __launch_bounds__(kThreads) void trunk_fwd_kernel(
+__global__ __launch_bounds__(kFwdThreads) void trunk_fwd_kernel(
     const uint8_t* __restrict__ frames, const float* __restrict__ w1,
     const float* __restrict__ b1, const float* __restrict__ w2,
     const float* __restrict__ b2, const float* __restrict__ w3,
@@ -64,100 +66,130 @@ __global__ __launch_bounds__(kThreads) void trunk_fwd_kernel(
     float* __restrict__ save2) {
   extern __shared__ unsigned char smem[];
   uint8_t* s_in = smem;
-  float* s_out1 = reinterpret_cast<float*>(smem + lds_in_bytes);
+  float* s_w1 = reinterpret_cast<float*>(smem + lds_in_bytes);
+  float* s_out1 = s_w1 + 32 * C * 64;
   float* s_out2 = s_out1 + 32 * H1 * W1;
 
   const int n = blockIdx.x;
   const int tid = threadIdx.x;
 
-  {  // Stage the u8 frame into LDS (word-wide where possible).
+  {  // Stage the u8 frame and the conv1 weights into LDS.
     const uint8_t* f = frames + (int64_t)n * C * H * W;
     const int total = C * H * W;
     const int words = total / 4;
     const uint32_t* f32p = reinterpret_cast<const uint32_t*>(f);
     uint32_t* s32p = reinterpret_cast<uint32_t*>(s_in);
-    for (int i = tid; i < words; i += kThreads) s32p[i] = f32p[i];
-    for (int i = words * 4 + tid; i < total; i += kThreads) s_in[i] = f[i];
+    for (int i = tid; i < words; i += kFwdThreads) s32p[i] = f32p[i];
+    for (int i = words * 4 + tid; i < total; i += kFwdThreads) s_in[i] = f[i];
+    const int w1n = 32 * C * 64;
+    for (int i = tid; i < w1n; i += kFwdThreads) s_w1[i] = w1[i];
   }
   __syncthreads();
 
-  {  // conv1: k8 s4, C -> 32, fused /255 normalize.
-    const int n1 = 32 * H1 * W1;
+  {  // conv1: k8 s4, C -> 32, fused /255. Two co per thread: the (shared)
+     // LDS frame reads feed two accumulator chains (ILP + half the reads).
     const int plane1 = H1 * W1;
-    for (int i = tid; i < n1; i += kThreads) {
-      const int co = i / plane1;
-      const int r = i - co * plane1;
+    const int npairs = 16 * plane1;
+    for (int i = tid; i < npairs; i += kFwdThreads) {
+      const int p = i / plane1;
+      const int r = i - p * plane1;
       const int oy = r / W1, ox = r - (r / W1) * W1;
-      float acc = 0.f;
-      const float* wco = w1 + (int64_t)co * C * 64;
+      const int co0 = 2 * p, co1 = 2 * p + 1;
+      float acc0 = 0.f, acc1 = 0.f;
+      const float* w0 = s_w1 + (int64_t)co0 * C * 64;
+      const float* w1r = s_w1 + (int64_t)co1 * C * 64;
       for (int ci = 0; ci < C; ++ci) {
         const uint8_t* in_c = s_in + ci * H * W + (oy * 4) * W + ox * 4;
-        const float* w_c = wco + ci * 64;
+        const float* wa = w0 + ci * 64;
+        const float* wb = w1r + ci * 64;
 #pragma unroll
         for (int ky = 0; ky < 8; ++ky) {
 #pragma unroll
           for (int kx = 0; kx < 8; ++kx) {
-            acc += (float)in_c[ky * W + kx] * w_c[ky * 8 + kx];
+            const float v = (float)in_c[ky * W + kx];
+            acc0 += v * wa[ky * 8 + kx];
+            acc1 += v * wb[ky * 8 + kx];
           }
         }
       }
-      const float v = fmaxf(acc * (1.f / 255.f) + b1[co], 0.f);
-      s_out1[i] = v;
-      if (save1 != nullptr) save1[(int64_t)n * n1 + i] = v;
+      const float v0 = fmaxf(acc0 * (1.f / 255.f) + b1[co0], 0.f);
+      const float v1 = fmaxf(acc1 * (1.f / 255.f) + b1[co1], 0.f);
+      s_out1[co0 * plane1 + r] = v0;
+      s_out1[co1 * plane1 + r] = v1;
+      if (save1 != nullptr) {
+        save1[(int64_t)n * 32 * plane1 + co0 * plane1 + r] = v0;
+        save1[(int64_t)n * 32 * plane1 + co1 * plane1 + r] = v1;
+      }
     }
   }
   __syncthreads();
 
-  {  // conv2: k4 s2, 32 -> 64.
-    const int n2 = 64 * H2 * W2;
+  {  // conv2: k4 s2, 32 -> 64, two co per thread.
     const int plane2 = H2 * W2;
-    for (int i = tid; i < n2; i += kThreads) {
-      const int co = i / plane2;
-      const int r = i - co * plane2;
+    const int npairs = 32 * plane2;
+    for (int i = tid; i < npairs; i += kFwdThreads) {
+      const int p = i / plane2;
+      const int r = i - p * plane2;
       const int oy = r / W2, ox = r - (r / W2) * W2;
-      float acc = b2[co];
-      const float* wco = w2 + (int64_t)co * 32 * 16;
+      const int co0 = 2 * p, co1 = 2 * p + 1;
+      float acc0 = b2[co0], acc1 = b2[co1];
+      const float* wa_base = w2 + (int64_t)co0 * 32 * 16;
+      const float* wb_base = w2 + (int64_t)co1 * 32 * 16;
       const float* in_base = s_out1 + (oy * 2) * W1 + ox * 2;
       for (int ci = 0; ci < 32; ++ci) {
         const float* in_c = in_base + ci * H1 * W1;
-        const float* w_c = wco + ci * 16;
+        const float* wa = wa_base + ci * 16;
+        const float* wb = wb_base + ci * 16;
 #pragma unroll
         for (int ky = 0; ky < 4; ++ky) {
 #pragma unroll
           for (int kx = 0; kx < 4; ++kx) {
-            acc += in_c[ky * W1 + kx] * w_c[ky * 4 + kx];
+            const float v = in_c[ky * W1 + kx];
+            acc0 += v * wa[ky * 4 + kx];
+            acc1 += v * wb[ky * 4 + kx];
           }
         }
       }
-      const float v = fmaxf(acc, 0.f);
-      s_out2[i] = v;
-      if (save2 != nullptr) save2[(int64_t)n * n2 + i] = v;
+      const float v0 = fmaxf(acc0, 0.f);
+      const float v1 = fmaxf(acc1, 0.f);
+      s_out2[co0 * plane2 + r] = v0;
+      s_out2[co1 * plane2 + r] = v1;
+      if (save2 != nullptr) {
+        save2[(int64_t)n * 64 * plane2 + co0 * plane2 + r] = v0;
+        save2[(int64_t)n * 64 * plane2 + co1 * plane2 + r] = v1;
+      }
     }
   }
   __syncthreads();
 
-  {  // conv3: k3 s1, 64 -> 64, straight to global (it's the fc input).
-    const int n3 = 64 * H3 * W3;
+  {  // conv3: k3 s1, 64 -> 64, two co per thread, straight to global.
     const int plane3 = H3 * W3;
-    for (int i = tid; i < n3; i += kThreads) {
-      const int co = i / plane3;
-      const int r = i - co * plane3;
+    const int npairs = 32 * plane3;
+    for (int i = tid; i < npairs; i += kFwdThreads) {
+      const int p = i / plane3;
+      const int r = i - p * plane3;
       const int oy = r / W3, ox = r - (r / W3) * W3;
-      float acc = b3[co];
-      const float* wco = w3 + (int64_t)co * 64 * 9;
+      const int co0 = 2 * p, co1 = 2 * p + 1;
+      float acc0 = b3[co0], acc1 = b3[co1];
+      const float* wa_base = w3 + (int64_t)co0 * 64 * 9;
+      const float* wb_base = w3 + (int64_t)co1 * 64 * 9;
       const float* in_base = s_out2 + oy * W2 + ox;
       for (int ci = 0; ci < 64; ++ci) {
         const float* in_c = in_base + ci * H2 * W2;
-        const float* w_c = wco + ci * 9;
+        const float* wa = wa_base + ci * 9;
+        const float* wb = wb_base + ci * 9;
 #pragma unroll
         for (int ky = 0; ky < 3; ++ky) {
 #pragma unroll
           for (int kx = 0; kx < 3; ++kx) {
-            acc += in_c[ky * W2 + kx] * w_c[ky * 3 + kx];
+            const float v = in_c[ky * W2 + kx];
+            acc0 += v * wa[ky * 3 + kx];
+            acc1 += v * wb[ky * 3 + kx];
           }
         }
       }
-      out3[(int64_t)n * n3 + i] = fmaxf(acc, 0.f);
+      out3[(int64_t)n * 64 * plane3 + co0 * plane3 + r] = fmaxf(acc0, 0.f);
+      out3[(int64_t)n * 64 * plane3 + co1 * plane3 + r] = fmaxf(acc1, 0.f);
     }
   }
 }
@@ -293,7 +325,7 @@ __global__ __launch_bounds__(kThreads) void conv_wgrad_kernel(
 }  // namespace
 
 bool atari_trunk_supported(int64_t C, int64_t H, int64_t W) {
-  if (H < 36 || W < 36 || C < 1 || C > 16) return false;
+  if (H < 36 || W < 36 || C < 1 || C > 8) return false;
   Geom g = make_geom(C, H, W);
   if (g.H3 < 1 || g.W3 < 1) return false;
   size_t bwd_lds = ((size_t)64 * g.H3 * g.W3 + (size_t)64 * g.H2 * g.W2) * 4;
@@ -331,7 +363,7 @@ std::vector<torch::Tensor> atari_trunk_fwd(
   auto w1c = w1.contiguous(); auto b1c = b1.contiguous();
   auto w2c = w2.contiguous(); auto b2c = b2.contiguous();
   auto w3c = w3.contiguous(); auto b3c = b3.contiguous();
-  hipLaunchKernelGGL(trunk_fwd_kernel, dim3(N), dim3(kThreads), g.lds_total,
+  hipLaunchKernelGGL(trunk_fwd_kernel, dim3(N), dim3(kFwdThreads), g.lds_total,
                      at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                      fr.data_ptr<uint8_t>(), w1c.data_ptr<float>(),
                      b1c.data_ptr<float>(), w2c.data_ptr<float>(),

@@ -113,15 +113,17 @@ class InferenceRunner {
     auto opts = torch::TensorOptions().device(device_);
 
     const int64_t C = frame.size(2), H = frame.size(3), W = frame.size(4);
-    torch::Tensor frames_p = torch::zeros({bp, C, H, W},
-                                          opts.dtype(torch::kUInt8));
+    torch::Tensor frames_p =
+        torch::empty({bp, C, H, W}, opts.dtype(torch::kUInt8));
     frames_p.narrow(0, 0, b).copy_(frame.reshape({b, C, H, W}),
                                    /*non_blocking=*/true);
+    if (bp > b) frames_p.narrow(0, b, bp - b).zero_();
 
-    torch::Tensor rew = torch::zeros({bp, 1}, opts.dtype(torch::kFloat32));
+    torch::Tensor rew = torch::empty({bp, 1}, opts.dtype(torch::kFloat32));
     rew.narrow(0, 0, b)
         .copy_(reward.to(opts.dtype(torch::kFloat32), true).reshape({b, 1}))
         .clamp_(-1, 1);
+    if (bp > b) rew.narrow(0, b, bp - b).zero_();
 
     torch::Tensor x;
     if (tbamd::atari_trunk_supported(C, H, W)) {
