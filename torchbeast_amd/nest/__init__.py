@@ -104,10 +104,14 @@ def front(nest):
     return f[0]
 
 
-try:  # Prefer the native implementation when built.
+try:  # Prefer the native implementation when the runtime extension is built.
     from torchbeast_amd.runtime import _tbruntime as _native
 
-    if hasattr(_native, "nest_map"):
-        pass  # Native overrides are wired up in runtime/__init__.py.
+    map = _native.map  # noqa: A001,F811
+    map_many = _native.map_many  # noqa: F811
+    map_many2 = _native.map_many2  # noqa: F811
+    flatten = _native.flatten  # noqa: F811
+    pack_as = _native.pack_as  # noqa: F811
+    front = _native.front  # noqa: F811
 except ImportError:
     pass
