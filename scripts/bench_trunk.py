@@ -4,7 +4,11 @@ GPU-only. Compares forward (inference + learner batch sizes) and the full
 fwd+bwd learner path. Run under gpurun; prints one line per config.
 """
 
+import os
+import sys
 import timeit
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.nn.functional as F
