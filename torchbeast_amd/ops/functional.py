@@ -306,6 +306,11 @@ def atari_trunk(frames, conv1, conv2, conv3):
     """
     if not frames.is_cuda or frames.dtype != torch.uint8:
         return None
+    # Measured crossover vs MIOpen on MI355X (profiles/trunkbench): the
+    # per-sample fused kernel wins below ~384 samples (single launch, no f32
+    # frame materialization); larger learner batches use library convs.
+    if frames.shape[0] > 384 and not os.environ.get("TBAMD_FORCE_TRUNK"):
+        return None
     ext = _ext_for(frames, "atari_trunk_fwd")
     if ext is None or not ext.atari_trunk_supported(*frames.shape[1:]):
         return None

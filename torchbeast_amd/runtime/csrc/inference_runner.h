@@ -126,7 +126,7 @@ class InferenceRunner {
     if (bp > b) rew.narrow(0, b, bp - b).zero_();
 
     torch::Tensor x;
-    if (tbamd::atari_trunk_supported(C, H, W)) {
+    if (bp <= 384 && tbamd::atari_trunk_supported(C, H, W)) {
       // Hand-written fused CDNA4 conv trunk: one kernel for the u8
       // normalize + 3 convs.
       x = tbamd::atari_trunk_fwd(frames_p, weights_[0], weights_[1],
