@@ -46,15 +46,15 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--batch_size", type=int, default=32)
     p.add_argument("--unroll_length", type=int, default=80)
-    p.add_argument("--actors", type=int, default=64,
-                   help="Env streams per GPU.")
+    p.add_argument("--actors", type=int, default=512,
+                   help="Env streams per GPU (tuned on MI355X; see profiles/).")
     p.add_argument("--model", default="shallow", choices=["shallow", "deep"])
     p.add_argument("--use_lstm", action="store_true")
     p.add_argument("--frame", default="4x84x84",
                    help="Synthetic frame shape CxHxW.")
     p.add_argument("--num_actions", type=int, default=6)
-    p.add_argument("--num_inference_threads", type=int, default=2)
-    p.add_argument("--inference_min_batch_size", type=int, default=1)
+    p.add_argument("--num_inference_threads", type=int, default=8)
+    p.add_argument("--inference_min_batch_size", type=int, default=128)
     p.add_argument("--inference_max_batch_size", type=int, default=512)
     p.add_argument("--inference_timeout_ms", type=int, default=5)
     p.add_argument("--episode_length", type=int, default=1000)
