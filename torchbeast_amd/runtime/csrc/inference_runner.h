@@ -17,6 +17,8 @@
 #include <torch/extension.h>
 
 #include <atomic>
+#include <mutex>
+#include <set>
 #include <thread>
 #include <vector>
 
@@ -96,6 +98,20 @@ class InferenceRunner {
 
   void serve(DynamicBatcher::Batch& batch, at::cuda::CUDAStream& stream) {
     TensorNest inputs = batch.get_inputs();
+    // Serialize the FIRST forward of each quantized batch size: concurrent
+    // MIOpen solution-finds for a brand-new conv shape across streams have
+    // been observed to fault; once found, the solution cache makes later
+    // serves safe to run fully in parallel.
+    const int64_t bq = (batch.size() + 63) / 64 * 64;
+    bool is_new;
+    {
+      std::lock_guard<std::mutex> g(warm_mu_);
+      is_new = warmed_sizes_.count(bq) == 0;
+    }
+    std::unique_lock<std::mutex> warm_lock;
+    if (is_new) {
+      warm_lock = std::unique_lock<std::mutex>(serve_mu_);
+    }
     // inputs = ((frame, reward, done, episode_step, episode_return), state)
     const auto& top = inputs.vector();
     const auto& env = top[0].vector();
@@ -224,8 +240,15 @@ class InferenceRunner {
 
     batches_.fetch_add(1, std::memory_order_relaxed);
     steps_.fetch_add(b, std::memory_order_relaxed);
+    if (warm_lock.owns_lock()) {
+      std::lock_guard<std::mutex> g(warm_mu_);
+      warmed_sizes_.insert(bq);
+    }
   }
 
+  std::mutex warm_mu_;
+  std::mutex serve_mu_;
+  std::set<int64_t> warmed_sizes_;
   std::shared_ptr<DynamicBatcher> batcher_;
   std::vector<torch::Tensor> weights_;
   const int64_t num_lstm_layers_;
