@@ -140,7 +140,7 @@ PYBIND11_MODULE(_tbruntime, m) {
 
   py::register_exception<ClosedQueue>(m, "ClosedBatchingQueue");
   py::register_exception<AsyncError>(m, "AsyncError");
-  py::register_exception<NestError>(m, "NestError");
+  py::register_exception<NestError>(m, "NestError", PyExc_ValueError);
 
   // ---- nest ops ----
   m.def("map", &nest_map, py::arg("function"), py::arg("nest"));
