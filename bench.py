@@ -176,7 +176,7 @@ def main():
     pool_thread.start()
 
     inference_runner = None
-    if args.model == "shallow" and use_cuda and not args.py_inference:
+    if use_cuda and not args.py_inference:
         inference_runner = pbl.make_inference_runner(
             actor_model, inference_batcher
         )

@@ -106,13 +106,25 @@ INFERENCE_STATS = {"batches": 0, "env_steps": 0}
 def make_inference_runner(actor_model, inference_batcher, greedy=False):
     """GIL-free C++ inference engine over the behavior model's parameters
     (views of the flat actor buffer, so the learner's one-copy weight sync
-    covers the runner too). Shallow AtariNet only; the deep ResNet uses the
-    Python inference path."""
+    covers the runner too). Supports the shallow AtariNet and the deep
+    IMPALA ResNet."""
     m = actor_model
-    weights = [
-        m.conv1.weight, m.conv1.bias,
-        m.conv2.weight, m.conv2.bias,
-        m.conv3.weight, m.conv3.bias,
+    if hasattr(m, "feat_extract"):  # deep ResNet
+        model_type = "deep"
+        weights = []
+        for section in m.feat_extract:
+            weights += [section.conv.weight, section.conv.bias]
+            for res in (section.res0, section.res1):
+                weights += [res.conv0.weight, res.conv0.bias,
+                            res.conv1.weight, res.conv1.bias]
+    else:
+        model_type = "shallow"
+        weights = [
+            m.conv1.weight, m.conv1.bias,
+            m.conv2.weight, m.conv2.bias,
+            m.conv3.weight, m.conv3.bias,
+        ]
+    weights += [
         m.fc.weight, m.fc.bias,
         m.policy.weight, m.policy.bias,
         m.baseline.weight, m.baseline.bias,
@@ -128,7 +140,8 @@ def make_inference_runner(actor_model, inference_batcher, greedy=False):
                 getattr(m.core, f"bias_hh_l{layer}"),
             ]
     return runtime._tbruntime.InferenceRunner(
-        inference_batcher, [w.detach() for w in weights], num_layers, greedy
+        inference_batcher, [w.detach() for w in weights], num_layers, greedy,
+        model_type,
     )
 
 _inference_stream = None
@@ -426,8 +439,7 @@ def train(flags):  # noqa: C901
         for i in range(1 if world_size > 1 else flags.num_learner_threads)
     ]
     use_cpp_inference = (
-        flags.model == "shallow"
-        and flags.actor_device.type == "cuda"
+        flags.actor_device.type == "cuda"
         and not getattr(flags, "py_inference", False)
     )
     inference_runner = None

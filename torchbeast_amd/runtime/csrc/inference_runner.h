@@ -29,18 +29,27 @@ namespace tbruntime {
 
 class InferenceRunner {
  public:
-  // weights order (shallow AtariNet, use_last_action=False):
+  // weights order:
+  //  shallow AtariNet (use_last_action=False):
   //   conv1.w, conv1.b, conv2.w, conv2.b, conv3.w, conv3.b,
-  //   fc.w, fc.b, policy.w, policy.b, baseline.w, baseline.b,
-  //   then per LSTM layer: w_ih, w_hh, b_ih, b_hh.
+  //   fc.w, fc.b, policy.w, policy.b, baseline.w, baseline.b
+  //  deep IMPALA ResNet: per section (x3): conv.w, conv.b,
+  //   res0.conv0.{w,b}, res0.conv1.{w,b}, res1.conv0.{w,b}, res1.conv1.{w,b}
+  //   then fc.w, fc.b, policy.w, policy.b, baseline.w, baseline.b
+  //  both: then per LSTM layer: w_ih, w_hh, b_ih, b_hh.
   InferenceRunner(std::shared_ptr<DynamicBatcher> batcher,
                   std::vector<torch::Tensor> weights, int64_t num_lstm_layers,
-                  bool greedy = false)
+                  bool greedy = false, std::string model_type = "shallow")
       : batcher_(std::move(batcher)),
         weights_(std::move(weights)),
         num_lstm_layers_(num_lstm_layers),
-        greedy_(greedy) {
-    TORCH_CHECK(weights_.size() >= 12, "need the 6 shallow-net layers");
+        greedy_(greedy),
+        deep_(model_type == "deep") {
+    const size_t trunk = deep_ ? 30 : 6;
+    head_base_ = trunk + 2;  // fc.w, fc.b come first after the trunk
+    lstm_base_ = trunk + 8;
+    TORCH_CHECK(weights_.size() >= lstm_base_ + 4 * (size_t)num_lstm_layers,
+                "runner weight list too short for model/lstm config");
     TORCH_CHECK(weights_[0].is_cuda(), "runner weights must be on the GPU");
     device_ = weights_[0].device();
   }
@@ -142,7 +151,23 @@ class InferenceRunner {
     if (bp > b) rew.narrow(0, b, bp - b).zero_();
 
     torch::Tensor x;
-    if (bp <= 384 && tbamd::atari_trunk_supported(C, H, W)) {
+    if (deep_) {
+      x = frames_p.to(torch::kFloat32).mul_(1.0f / 255.0f);
+      size_t wi = 0;
+      for (int sec = 0; sec < 3; ++sec) {
+        x = at::conv2d(x, weights_[wi], weights_[wi + 1], 1, 1);
+        wi += 2;
+        x = at::max_pool2d(x, 3, 2, 1);
+        for (int res = 0; res < 2; ++res) {
+          torch::Tensor y = at::conv2d(at::relu(x), weights_[wi],
+                                       weights_[wi + 1], 1, 1);
+          y = at::conv2d(y.relu_(), weights_[wi + 2], weights_[wi + 3], 1, 1);
+          wi += 4;
+          x = x + y;
+        }
+      }
+      x = at::relu(x).reshape({bp, -1});
+    } else if (bp <= 384 && tbamd::atari_trunk_supported(C, H, W)) {
       // Hand-written fused CDNA4 conv trunk: one kernel for the u8
       // normalize + 3 convs.
       x = tbamd::atari_trunk_fwd(frames_p, weights_[0], weights_[1],
@@ -155,7 +180,8 @@ class InferenceRunner {
       x = at::conv2d(x, weights_[4], weights_[5], 1).relu_();
       x = x.reshape({bp, -1});
     }
-    x = at::linear(x, weights_[6], weights_[7]).relu_();
+    x = at::linear(x, weights_[head_base_ - 2], weights_[head_base_ - 1])
+            .relu_();
     torch::Tensor core = at::cat({x, rew}, 1);
 
     std::vector<torch::Tensor> new_state_gpu;
@@ -176,10 +202,10 @@ class InferenceRunner {
       torch::Tensor layer_in = core;
       std::vector<torch::Tensor> hs, cs;
       for (int64_t l = 0; l < num_lstm_layers_; ++l) {
-        const auto& w_ih = weights_[12 + 4 * l];
-        const auto& w_hh = weights_[12 + 4 * l + 1];
-        const auto& b_ih = weights_[12 + 4 * l + 2];
-        const auto& b_hh = weights_[12 + 4 * l + 3];
+        const auto& w_ih = weights_[lstm_base_ + 4 * l];
+        const auto& w_hh = weights_[lstm_base_ + 4 * l + 1];
+        const auto& b_ih = weights_[lstm_base_ + 4 * l + 2];
+        const auto& b_hh = weights_[lstm_base_ + 4 * l + 3];
         torch::Tensor gates = at::addmm(b_ih + b_hh, layer_in, w_ih.t())
                                   .addmm_(h[l], w_hh.t());
         auto chunks = gates.chunk(4, 1);
@@ -199,10 +225,12 @@ class InferenceRunner {
     }
 
     torch::Tensor logits =
-        at::linear(core, weights_[8], weights_[9]).narrow(0, 0, b);
+        at::linear(core, weights_[head_base_], weights_[head_base_ + 1])
+            .narrow(0, 0, b);
     torch::Tensor baseline =
-        at::linear(core, weights_[10], weights_[11]).narrow(0, 0, b).reshape(
-            {b});
+        at::linear(core, weights_[head_base_ + 2], weights_[head_base_ + 3])
+            .narrow(0, 0, b)
+            .reshape({b});
     torch::Tensor action;
     if (greedy_) {
       action = at::argmax(logits, -1);
@@ -253,6 +281,9 @@ class InferenceRunner {
   std::vector<torch::Tensor> weights_;
   const int64_t num_lstm_layers_;
   const bool greedy_;
+  bool deep_ = false;
+  size_t head_base_ = 8;
+  size_t lstm_base_ = 12;
   torch::Device device_ = torch::kCPU;
   std::atomic<bool> running_{false};
   std::atomic<int64_t> batches_{0};
