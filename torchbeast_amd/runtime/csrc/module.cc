@@ -248,9 +248,10 @@ PYBIND11_MODULE(_tbruntime, m) {
   py::class_<InferenceRunner, std::shared_ptr<InferenceRunner>>(
       m, "InferenceRunner")
       .def(py::init<std::shared_ptr<DynamicBatcher>,
-                    std::vector<torch::Tensor>, int64_t, bool>(),
+                    std::vector<torch::Tensor>, int64_t, bool, std::string>(),
            py::arg("inference_batcher"), py::arg("weights"),
-           py::arg("num_lstm_layers") = 0, py::arg("greedy") = false)
+           py::arg("num_lstm_layers") = 0, py::arg("greedy") = false,
+           py::arg("model_type") = "shallow")
       .def("start", &InferenceRunner::start, py::arg("num_threads") = 2,
            py::call_guard<py::gil_scoped_release>())
       .def("stop", &InferenceRunner::stop,
