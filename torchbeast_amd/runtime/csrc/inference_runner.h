@@ -47,7 +47,7 @@ class InferenceRunner {
         deep_(model_type == "deep") {
     const size_t trunk = deep_ ? 30 : 6;
     head_base_ = trunk + 2;  // fc.w, fc.b come first after the trunk
-    lstm_base_ = trunk + 8;
+    lstm_base_ = trunk + 6;  // fc(2) + policy(2) + baseline(2)
     TORCH_CHECK(weights_.size() >= lstm_base_ + 4 * (size_t)num_lstm_layers,
                 "runner weight list too short for model/lstm config");
     TORCH_CHECK(weights_[0].is_cuda(), "runner weights must be on the GPU");
