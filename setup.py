@@ -18,6 +18,11 @@ os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
 
+# Sanitizer flavors for the C++ runtime (race/heap checking of the queue and
+# actor-pool concurrency): TBAMD_SANITIZE=thread|address.
+_sanitize = os.environ.get("TBAMD_SANITIZE")
+_san_flags = [f"-fsanitize={_sanitize}", "-fno-omit-frame-pointer"] if _sanitize else []
+
 ext_modules = [
     # CUDAExtension (ROCm) so the runtime can drive HIP streams/ATen-GPU
     # ops from C++ (GIL-free inference engine).
@@ -28,9 +33,10 @@ ext_modules = [
             "torchbeast_amd/ops/hip/atari_trunk.hip",
         ],
         extra_compile_args={
-            "cxx": ["-O3", "-std=c++17", "-pthread"],
+            "cxx": ["-O3", "-std=c++17", "-pthread"] + _san_flags,
             "nvcc": ["-O3", "-std=c++17"],
         },
+        extra_link_args=_san_flags,
     )
 ]
 
