@@ -286,6 +286,12 @@ class BatchingQueue {
       at::cuda::CUDAEvent ev;
       ev.record(*copy_stream_);
       ev.block(current);
+      // The batch blocks were allocated on the copy stream but are consumed
+      // (and eventually freed) on the caller's stream: tell the caching
+      // allocator, or it may recycle them while the consumer still reads.
+      batched.for_each([&current](const torch::Tensor& t) {
+        t.record_stream(current);
+      });
     } else {
       batched = batch_nests(ptrs, batch_dim_);
     }
