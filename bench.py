@@ -117,6 +117,10 @@ def main():
     args = parse_args()
     rank, world_size, local_rank = tbddp.maybe_init_distributed()
 
+    # Distinct RNG per rank (sampling, init noise); weights are broadcast
+    # from rank 0 afterwards, so replicas still start identical.
+    torch.manual_seed(4242 + rank * 977)
+
     use_cuda = torch.cuda.is_available()
     device = torch.device("cuda", local_rank) if use_cuda else torch.device("cpu")
     if use_cuda:
@@ -171,6 +175,7 @@ def main():
         inference_batcher=inference_batcher,
         env_server_addresses=addresses,
         initial_agent_state=initial_agent_state,
+        seed_base=rank * args.actors,  # distinct env streams per rank
     )
     pool_thread = threading.Thread(target=pool.run, daemon=True)
     pool_thread.start()

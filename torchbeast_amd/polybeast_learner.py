@@ -308,6 +308,7 @@ def learn(
 def train(flags):  # noqa: C901
     rank, world_size, local_rank = tbddp.maybe_init_distributed()
     is_leader = rank == 0
+    torch.manual_seed(4242 + rank * 977)
 
     if flags.xpid is None:
         flags.xpid = "polybeast-%s" % time.strftime("%Y%m%d-%H%M%S")
@@ -412,6 +413,7 @@ def train(flags):  # noqa: C901
         inference_batcher=inference_batcher,
         env_server_addresses=addresses,
         initial_agent_state=initial_agent_state,
+        seed_base=rank * flags.num_actors,
     )
 
     def run_pool():

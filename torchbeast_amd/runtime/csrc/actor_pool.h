@@ -25,12 +25,13 @@ class ActorPool {
   ActorPool(int64_t unroll_length, std::shared_ptr<BatchingQueue> learner_queue,
             std::shared_ptr<DynamicBatcher> inference_batcher,
             std::vector<std::string> env_server_addresses,
-            TensorNest initial_agent_state)
+            TensorNest initial_agent_state, int64_t seed_base = 0)
       : unroll_length_(unroll_length),
         learner_queue_(std::move(learner_queue)),
         inference_batcher_(std::move(inference_batcher)),
         addresses_(std::move(env_server_addresses)),
-        initial_agent_state_(std::move(initial_agent_state)) {
+        initial_agent_state_(std::move(initial_agent_state)),
+        seed_base_(seed_base) {
     if (unroll_length_ < 1) {
       throw std::invalid_argument("unroll_length must be >= 1");
     }
@@ -62,7 +63,7 @@ class ActorPool {
 
  private:
   void loop(const std::string& address, uint64_t seed) {
-    auto env = make_env_connection(address, seed + 1);
+    auto env = make_env_connection(address, seed_base_ + seed + 1);
 
     TensorNest env_outputs = env->initial();
     TensorNest agent_state = initial_agent_state_;
@@ -117,6 +118,7 @@ class ActorPool {
   std::shared_ptr<DynamicBatcher> inference_batcher_;
   std::vector<std::string> addresses_;
   TensorNest initial_agent_state_;
+  const int64_t seed_base_;
   std::atomic<uint64_t> step_count_{0};
 };
 
