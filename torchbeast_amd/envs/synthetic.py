@@ -64,6 +64,38 @@ class SyntheticAtariEnv:
         pass
 
 
+class BanditEnv:
+    """Learnability fixture: constant observation, reward 1 for the target
+    action and 0 otherwise. A correct IMPALA implementation drives the
+    policy toward the target within a few thousand steps; tests assert the
+    mean episode return rises (the framework-level analogue of the
+    reference's 'learning curves equivalent' claim)."""
+
+    def __init__(self, shape=(4, 36, 36), num_actions=4, target_action=2,
+                 episode_length=20):
+        self.shape = tuple(shape)
+        self.num_actions = num_actions
+        self.target_action = target_action
+        self.episode_length = episode_length
+        self._t = 0
+        self._obs = np.full(self.shape, 128, dtype=np.uint8)
+
+    def reset(self):
+        self._t = 0
+        return self._obs
+
+    def step(self, action):
+        self._t += 1
+        reward = 1.0 if int(action) == self.target_action else 0.0
+        done = self._t >= self.episode_length
+        if done:
+            self._t = 0
+        return self._obs, reward, done, {}
+
+    def close(self):
+        pass
+
+
 class CountingEnv:
     """Observation = global step counter; episodes of fixed length.
 
