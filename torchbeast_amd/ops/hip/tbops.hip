@@ -373,16 +373,19 @@ torch::Tensor policy_sample(torch::Tensor logits, bool greedy, int64_t seed) {
 // Done-masked LSTM unroll (persistent cooperative kernel per layer).
 // ---------------------------------------------------------------------------
 
-// Forward, one layer. Grid-wide sync between the gate phase (threads cover
-// B*4H GEMV outputs) and the state phase (threads cover B*H elements).
-// Stash layout (all [T,B,*]): gates post-activation (4H), hm, cm, c.
-__global__ void lstm_fwd_kernel(
+// Forward, one layer, persistent cooperative kernel. Each workgroup OWNS a
+// slice of W_hh rows staged in LDS for the whole unroll (one load instead of
+// T re-reads from L3 — W_hh is ~4 MB for hidden 519, larger than one XCD's
+// L2). Per step: gate phase (each WG computes its gate slice for all B) ->
+// grid.sync -> state-update phase (grid-stride over B*H, which also writes
+// the next step's done-masked state) -> grid.sync.
+__global__ __launch_bounds__(256) void lstm_fwd_kernel(
     const float* __restrict__ precomp,  // [T,B,4H] = x@W_ih^T + b_ih + b_hh
     const float* __restrict__ w_hh,     // [4H,H]
     const float* __restrict__ notdone,  // [T,B]
     const float* __restrict__ h0,       // [B,H]
     const float* __restrict__ c0,       // [B,H]
-    int T, int B, int H,
+    int T, int B, int H, int j_slice,
     float* __restrict__ out,    // [T,B,H]
     float* __restrict__ gates,  // [T,B,4H] post-activation
     float* __restrict__ hm,     // [T,B,H] masked h_{t-1}
@@ -391,15 +394,24 @@ __global__ void lstm_fwd_kernel(
     float* __restrict__ hT,     // [B,H]
     float* __restrict__ cT) {   // [B,H]
   cg::grid_group grid = cg::this_grid();
+  extern __shared__ float s_w[];  // [j_slice, H]
   const int64_t tid = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   const int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
   const int64_t BH = (int64_t)B * H;
-  const int64_t BG = (int64_t)B * 4 * H;
+  const int G4 = 4 * H;
 
-  // Step -1: initialize hm/cm for t=0.
+  const int j0 = blockIdx.x * j_slice;
+  const int jn = min(j_slice, G4 - j0);
+
+  // Stage this workgroup's W_hh rows into LDS (once for the whole unroll).
+  for (int i = threadIdx.x; i < jn * H; i += blockDim.x) {
+    s_w[i] = w_hh[(int64_t)j0 * H + i];
+  }
+
+  // Initialize hm/cm for t=0.
   for (int64_t i = tid; i < BH; i += nthreads) {
     const int b = i / H;
-    const float nd = notdone[b];  // notdone[0, b]
+    const float nd = notdone[b];
     hm[i] = nd * h0[i];
     cm[i] = nd * c0[i];
   }
@@ -408,25 +420,24 @@ __global__ void lstm_fwd_kernel(
   for (int t = 0; t < T; ++t) {
     const float* hm_t = hm + (int64_t)t * BH;
     const float* cm_t = cm + (int64_t)t * BH;
-    float* gates_t = gates + (int64_t)t * BG;
-    const float* pre_t = precomp + (int64_t)t * BG;
+    float* gates_t = gates + (int64_t)t * (int64_t)B * G4;
+    const float* pre_t = precomp + (int64_t)t * (int64_t)B * G4;
 
-    // Gate GEMV: gates[b, j] = pre[b, j] + dot(hm[b, :], w_hh[j, :]).
-    for (int64_t i = tid; i < BG; i += nthreads) {
-      const int b = i / (4 * H);
-      const int j = i % (4 * H);
+    // Gate phase: this WG computes gates[b, j0..j0+jn) for all b.
+    for (int i = threadIdx.x; i < jn * B; i += blockDim.x) {
+      const int b = i / jn;           // lanes iterate j fastest: the hm row
+      const int jl = i - (i / jn) * jn;  // is shared across the wave.
       const float* hrow = hm_t + (int64_t)b * H;
-      const float* wrow = w_hh + (int64_t)j * H;
-      float acc = pre_t[i];
+      const float* wrow = s_w + (int64_t)jl * H;
+      float acc = pre_t[(int64_t)b * G4 + j0 + jl];
       for (int h = 0; h < H; ++h) acc += hrow[h] * wrow[h];
-      // Activations: i, f, o sigmoid; g tanh (PyTorch gate order i,f,g,o).
-      const int gate = j / H;
+      const int gate = (j0 + jl) / H;
       if (gate == 2) {
         acc = tanhf(acc);
       } else {
         acc = 1.f / (1.f + __expf(-acc));
       }
-      gates_t[i] = acc;
+      gates_t[(int64_t)b * G4 + j0 + jl] = acc;
     }
     grid.sync();
 
@@ -435,8 +446,8 @@ __global__ void lstm_fwd_kernel(
     float* c_t = c_out + (int64_t)t * BH;
     for (int64_t i = tid; i < BH; i += nthreads) {
       const int b = i / H;
-      const int h = i % H;
-      const float* g4 = gates_t + (int64_t)b * 4 * H;
+      const int h = i - (int64_t)(i / H) * H;
+      const float* g4 = gates_t + (int64_t)b * G4;
       const float gi = g4[h];
       const float gf = g4[H + h];
       const float gg = g4[2 * H + h];
@@ -459,26 +470,36 @@ __global__ void lstm_fwd_kernel(
 }
 
 // Backward, one layer: computes pre-activation gate grads dgates [T,B,4H]
-// and the carried dh/dc (written to dh0/dc0 at the end). Weight/input grads
-// are batched GEMMs done by the host with rocBLAS.
-__global__ void lstm_bwd_kernel(
+// and the carried dh/dc. Phase 1 (elementwise over B*H) produces dgates;
+// phase 2 computes dh_{t-1} = dgates_t @ W_hh using a per-workgroup LDS
+// slice of W_hh^T held across all T steps (same persistent-weights design
+// as the forward). Weight/input grads are batched rocBLAS GEMMs on the host.
+__global__ __launch_bounds__(256) void lstm_bwd_kernel(
     const float* __restrict__ gates,    // [T,B,4H] post-activation
     const float* __restrict__ cm,       // [T,B,H]
     const float* __restrict__ c_out,    // [T,B,H]
-    const float* __restrict__ w_hh,     // [4H,H]
+    const float* __restrict__ w_hh_t,   // [H,4H] = W_hh^T (contiguous)
     const float* __restrict__ notdone,  // [T,B]
     const float* __restrict__ d_out,    // [T,B,H]
     const float* __restrict__ d_hT,     // [B,H]
     const float* __restrict__ d_cT,     // [B,H]
-    int T, int B, int H,
+    int T, int B, int H, int h_slice,
     float* __restrict__ dgates,  // [T,B,4H] PRE-activation grads
     float* __restrict__ dh,      // [B,H] workspace, ends as d_h0
     float* __restrict__ dc) {    // [B,H] workspace, ends as d_c0
   cg::grid_group grid = cg::this_grid();
+  extern __shared__ float s_wt[];  // [h_slice, 4H]
   const int64_t tid = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   const int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
   const int64_t BH = (int64_t)B * H;
-  const int64_t BG = (int64_t)B * 4 * H;
+  const int G4 = 4 * H;
+
+  const int hh0 = blockIdx.x * h_slice;
+  const int hn = min(h_slice, H - hh0);
+
+  for (int i = threadIdx.x; i < hn * G4; i += blockDim.x) {
+    s_wt[i] = w_hh_t[(int64_t)hh0 * G4 + i];
+  }
 
   for (int64_t i = tid; i < BH; i += nthreads) {
     dh[i] = d_hT[i];
@@ -487,16 +508,16 @@ __global__ void lstm_bwd_kernel(
   grid.sync();
 
   for (int t = T - 1; t >= 0; --t) {
-    const float* gates_t = gates + (int64_t)t * BG;
+    const float* gates_t = gates + (int64_t)t * (int64_t)B * G4;
     const float* cm_t = cm + (int64_t)t * BH;
     const float* c_t = c_out + (int64_t)t * BH;
-    float* dgates_t = dgates + (int64_t)t * BG;
+    float* dgates_t = dgates + (int64_t)t * (int64_t)B * G4;
 
-    // Phase 1: per-(b,h) gate gradients; dc becomes dcm (pre-mask).
+    // Phase 1: per-(b,h) gate gradients; dc becomes masked d c_{t-1}.
     for (int64_t i = tid; i < BH; i += nthreads) {
       const int b = i / H;
-      const int h = i % H;
-      const float* g4 = gates_t + (int64_t)b * 4 * H;
+      const int h = i - (int64_t)(i / H) * H;
+      const float* g4 = gates_t + (int64_t)b * G4;
       const float gi = g4[h];
       const float gf = g4[H + h];
       const float gg = g4[2 * H + h];
@@ -511,27 +532,27 @@ __global__ void lstm_bwd_kernel(
       const float d_gf = dc_t * cm_t[i];
       const float d_gg = dc_t * gi;
 
-      float* dg4 = dgates_t + (int64_t)b * 4 * H;
+      float* dg4 = dgates_t + (int64_t)b * G4;
       dg4[h] = d_gi * gi * (1.f - gi);
       dg4[H + h] = d_gf * gf * (1.f - gf);
       dg4[2 * H + h] = d_gg * (1.f - gg * gg);
       dg4[3 * H + h] = d_go * go * (1.f - go);
 
-      // d cm = dc * f; then mask to get d c_{t-1}.
       const float nd = notdone[(int64_t)t * B + b];
       dc[i] = nd * dc_t * gf;
     }
     grid.sync();
 
-    // Phase 2: dh_{t-1} = notdone_t * (dgates_t @ W_hh)[b, h].
-    for (int64_t i = tid; i < BH; i += nthreads) {
-      const int b = i / H;
-      const int h = i % H;
-      const float* dg4 = dgates_t + (int64_t)b * 4 * H;
+    // Phase 2: dh_{t-1}[b, h] for this WG's h-slice, from the LDS W_hh^T.
+    for (int i = threadIdx.x; i < hn * B; i += blockDim.x) {
+      const int b = i / hn;
+      const int hl = i - (i / hn) * hn;
+      const float* dg4 = dgates_t + (int64_t)b * G4;
+      const float* wrow = s_wt + (int64_t)hl * G4;
       float acc = 0.f;
-      for (int j = 0; j < 4 * H; ++j) acc += dg4[j] * w_hh[(int64_t)j * H + h];
+      for (int j = 0; j < G4; ++j) acc += dg4[j] * wrow[j];
       const float nd = notdone[(int64_t)t * B + b];
-      dh[i] = nd * acc;
+      dh[(int64_t)b * H + hh0 + hl] = nd * acc;
     }
     grid.sync();
   }
@@ -569,9 +590,16 @@ std::vector<torch::Tensor> lstm_unroll_fwd(
   torch::Tensor out;
 
   const int threads = 256;
-  static int max_grid = 0;
-  if (max_grid == 0) {
-    max_grid = coop_grid_size((const void*)lstm_fwd_kernel, threads, 0);
+  // Workgroup count: each WG owns a W_hh row-slice held in LDS. Cap the
+  // per-WG slice at ~48 KiB so several blocks fit per CU.
+  const int64_t w_bytes = (int64_t)4 * H * H * 4;
+  int grid = std::max<int64_t>(128, ceil_div(w_bytes, 48 * 1024));
+  const int j_slice = ceil_div((int64_t)4 * H, grid);
+  const size_t fwd_lds = (size_t)j_slice * H * sizeof(float);
+  TORCH_CHECK(fwd_lds <= 64 * 1024, "lstm hidden size too large for LDS slices");
+  {
+    int cap = coop_grid_size((const void*)lstm_fwd_kernel, threads, fwd_lds);
+    TORCH_CHECK(grid <= cap, "lstm fwd grid exceeds cooperative capacity");
   }
 
   for (int l = 0; l < L; ++l) {
@@ -594,10 +622,6 @@ std::vector<torch::Tensor> lstm_unroll_fwd(
     auto h0_l = h0[l].contiguous();
     auto c0_l = c0[l].contiguous();
 
-    const int64_t work = (int64_t)B * 4 * H;
-    const int grid =
-        std::min<int64_t>(max_grid, ceil_div(work, threads));
-
     const float* precomp_p = precomp.data_ptr<float>();
     const float* w_hh_p = w_hh.data_ptr<float>();
     const float* notdone_p = notdone_f.data_ptr<float>();
@@ -610,13 +634,14 @@ std::vector<torch::Tensor> lstm_unroll_fwd(
     float* c_out_p = c_out.data_ptr<float>();
     float* hT_p = hT[l].data_ptr<float>();
     float* cT_p = cT[l].data_ptr<float>();
-    int T_ = T, B_ = B, H_ = H;
+    int T_ = T, B_ = B, H_ = H, j_slice_ = j_slice;
     void* args[] = {&precomp_p, &w_hh_p, &notdone_p, &h0_p, &c0_p,
-                    &T_,        &B_,     &H_,        &out_p, &gates_p,
-                    &hm_p,      &cm_p,   &c_out_p,   &hT_p,  &cT_p};
+                    &T_,        &B_,     &H_,        &j_slice_,
+                    &out_p,     &gates_p, &hm_p,     &cm_p,
+                    &c_out_p,   &hT_p,   &cT_p};
     DEVCHECK(hipLaunchCooperativeKernel((const void*)lstm_fwd_kernel,
-                                        dim3(grid), dim3(threads), args, 0,
-                                        stream));
+                                        dim3(grid), dim3(threads), args,
+                                        fwd_lds, stream));
 
     stash.push_back(layer_in);
     stash.push_back(gates);
@@ -649,9 +674,14 @@ std::vector<torch::Tensor> lstm_unroll_bwd(
   std::vector<torch::Tensor> d_weights(4 * L);
 
   const int threads = 256;
-  static int max_grid = 0;
-  if (max_grid == 0) {
-    max_grid = coop_grid_size((const void*)lstm_bwd_kernel, threads, 0);
+  const int64_t w_bytes = (int64_t)4 * H * H * 4;
+  int grid = std::max<int64_t>(128, ceil_div(w_bytes, 48 * 1024));
+  const int h_slice = ceil_div((int64_t)H, grid);
+  const size_t bwd_lds = (size_t)h_slice * 4 * H * sizeof(float);
+  TORCH_CHECK(bwd_lds <= 64 * 1024, "lstm hidden size too large for LDS slices");
+  {
+    int cap = coop_grid_size((const void*)lstm_bwd_kernel, threads, bwd_lds);
+    TORCH_CHECK(grid <= cap, "lstm bwd grid exceeds cooperative capacity");
   }
 
   torch::Tensor d_out = d_out_top.contiguous();
@@ -663,6 +693,7 @@ std::vector<torch::Tensor> lstm_unroll_bwd(
     auto c_out = stash[5 * l + 4];
     auto w_ih = flat_weights[4 * l].contiguous();
     auto w_hh = flat_weights[4 * l + 1].contiguous();
+    auto w_hh_t = w_hh.t().contiguous();  // [H, 4H] for the bwd LDS slices
 
     auto dgates = torch::empty({T, B, 4 * H}, opts);
     auto dh = torch::empty({B, H}, opts);
@@ -670,13 +701,10 @@ std::vector<torch::Tensor> lstm_unroll_bwd(
     auto d_hT_l = d_hT[l].contiguous();
     auto d_cT_l = d_cT[l].contiguous();
 
-    const int grid =
-        std::min<int64_t>(max_grid, ceil_div((int64_t)B * H, threads));
-
     const float* gates_p = gates.data_ptr<float>();
     const float* cm_p = cm.data_ptr<float>();
     const float* c_out_p = c_out.data_ptr<float>();
-    const float* w_hh_p = w_hh.data_ptr<float>();
+    const float* w_hh_t_p = w_hh_t.data_ptr<float>();
     const float* notdone_p = notdone_f.data_ptr<float>();
     const float* d_out_p = d_out.data_ptr<float>();
     const float* d_hT_p = d_hT_l.data_ptr<float>();
@@ -684,13 +712,13 @@ std::vector<torch::Tensor> lstm_unroll_bwd(
     float* dgates_p = dgates.data_ptr<float>();
     float* dh_p = dh.data_ptr<float>();
     float* dc_p = dc.data_ptr<float>();
-    int T_ = T, B_ = B, H_ = H;
-    void* args[] = {&gates_p, &cm_p, &c_out_p, &w_hh_p, &notdone_p,
-                    &d_out_p, &d_hT_p, &d_cT_p, &T_, &B_, &H_,
+    int T_ = T, B_ = B, H_ = H, h_slice_ = h_slice;
+    void* args[] = {&gates_p, &cm_p, &c_out_p, &w_hh_t_p, &notdone_p,
+                    &d_out_p, &d_hT_p, &d_cT_p, &T_, &B_, &H_, &h_slice_,
                     &dgates_p, &dh_p, &dc_p};
     DEVCHECK(hipLaunchCooperativeKernel((const void*)lstm_bwd_kernel,
-                                        dim3(grid), dim3(threads), args, 0,
-                                        stream));
+                                        dim3(grid), dim3(threads), args,
+                                        bwd_lds, stream));
 
     // Weight/input grads: plain GEMMs (rocBLAS).
     auto dg2 = dgates.reshape({(int64_t)T * B, 4 * H});
