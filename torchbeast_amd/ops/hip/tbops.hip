@@ -373,19 +373,19 @@ torch::Tensor policy_sample(torch::Tensor logits, bool greedy, int64_t seed) {
 // Done-masked LSTM unroll (persistent cooperative kernel per layer).
 // ---------------------------------------------------------------------------
 
-// Forward, one layer. One workgroup PER BATCH ELEMENT: the recurrences of
-// different batch rows are independent, so there is no cross-workgroup
-// dependency and no grid-wide sync (measured ~100 us per grid.sync() across
-// MI355X's 8 XCDs — the earlier cooperative design spent most of its time
-// there). h/c live in LDS for the whole unroll; W_hh^T is streamed
-// lane-coalesced from L2/L3 (each lane owns output columns j, j+256, ...).
+// Forward, one layer, persistent cooperative kernel. Each workgroup OWNS a
+// slice of W_hh rows staged in LDS for the whole unroll (one load instead of
+// T re-reads from L3 — W_hh is ~4 MB for hidden 519, larger than one XCD's
+// L2). Per step: gate phase (each WG computes its gate slice for all B) ->
+// grid.sync -> state-update phase (grid-stride over B*H, which also writes
+// the next step's done-masked state) -> grid.sync.
 __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     const float* __restrict__ precomp,  // [T,B,4H] = x@W_ih^T + b_ih + b_hh
-    const float* __restrict__ w_hh_t,   // [H,4H] = W_hh^T (contiguous)
+    const float* __restrict__ w_hh,     // [4H,H]
     const float* __restrict__ notdone,  // [T,B]
     const float* __restrict__ h0,       // [B,H]
     const float* __restrict__ c0,       // [B,H]
-    int T, int B, int H,
+    int T, int B, int H, int j_slice,
     float* __restrict__ out,    // [T,B,H]
     float* __restrict__ gates,  // [T,B,4H] post-activation
     float* __restrict__ hm,     // [T,B,H] masked h_{t-1}
@@ -393,187 +393,168 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     float* __restrict__ c_out,  // [T,B,H]
     float* __restrict__ hT,     // [B,H]
     float* __restrict__ cT) {   // [B,H]
-  extern __shared__ float smem[];
-  float* s_h = smem;          // [H] current (masked) h
-  float* s_c = s_h + H;       // [H] current (masked) c
-  float* s_g = s_c + H;       // [4H] gates of this step
-
-  const int b = blockIdx.x;
-  const int tid = threadIdx.x;
-  const int G4 = 4 * H;
+  cg::grid_group grid = cg::this_grid();
+  extern __shared__ float s_w[];  // [j_slice, H]
+  const int64_t tid = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  const int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
   const int64_t BH = (int64_t)B * H;
+  const int G4 = 4 * H;
 
-  for (int i = tid; i < H; i += blockDim.x) {
-    s_h[i] = h0[(int64_t)b * H + i];
-    s_c[i] = c0[(int64_t)b * H + i];
+  const int j0 = blockIdx.x * j_slice;
+  const int jn = min(j_slice, G4 - j0);
+
+  // Stage this workgroup's W_hh rows into LDS (once for the whole unroll).
+  for (int i = threadIdx.x; i < jn * H; i += blockDim.x) {
+    s_w[i] = w_hh[(int64_t)j0 * H + i];
   }
-  __syncthreads();
 
-  // Each lane owns gate columns j = tid + k*256.
-  constexpr int kMaxJ = 16;  // supports 4H up to 4096 (H <= 1024)
-  const int nj = (G4 - tid + 255) / 256;
+  // Initialize hm/cm for t=0.
+  for (int64_t i = tid; i < BH; i += nthreads) {
+    const int b = i / H;
+    const float nd = notdone[b];
+    hm[i] = nd * h0[i];
+    cm[i] = nd * c0[i];
+  }
+  grid.sync();
 
   for (int t = 0; t < T; ++t) {
-    const float nd = notdone[(int64_t)t * B + b];
+    const float* hm_t = hm + (int64_t)t * BH;
+    const float* cm_t = cm + (int64_t)t * BH;
+    float* gates_t = gates + (int64_t)t * (int64_t)B * G4;
+    const float* pre_t = precomp + (int64_t)t * (int64_t)B * G4;
 
-    // Mask state (and stash hm/cm for the backward pass).
-    for (int i = tid; i < H; i += blockDim.x) {
-      const float hv = nd * s_h[i];
-      const float cv = nd * s_c[i];
-      s_h[i] = hv;
-      s_c[i] = cv;
-      hm[(int64_t)t * BH + (int64_t)b * H + i] = hv;
-      cm[(int64_t)t * BH + (int64_t)b * H + i] = cv;
+    // Gate phase: this WG computes gates[b, j0..j0+jn) for all b.
+    for (int i = threadIdx.x; i < jn * B; i += blockDim.x) {
+      const int b = i / jn;           // lanes iterate j fastest: the hm row
+      const int jl = i - (i / jn) * jn;  // is shared across the wave.
+      const float* hrow = hm_t + (int64_t)b * H;
+      const float* wrow = s_w + (int64_t)jl * H;
+      float acc = pre_t[(int64_t)b * G4 + j0 + jl];
+      for (int h = 0; h < H; ++h) acc += hrow[h] * wrow[h];
+      const int gate = (j0 + jl) / H;
+      if (gate == 2) {
+        acc = tanhf(acc);
+      } else {
+        acc = 1.f / (1.f + __expf(-acc));
+      }
+      gates_t[(int64_t)b * G4 + j0 + jl] = acc;
     }
-    __syncthreads();
+    grid.sync();
 
-    // Gates: acc[k] for columns tid + 256k; W_hh^T rows read coalesced.
-    // kMaxJ-bounded unrolled loops with guards: a runtime trip count would
-    // push acc[] to scratch memory.
-    float acc[kMaxJ];
-#pragma unroll
-    for (int k = 0; k < kMaxJ; ++k) acc[k] = 0.f;
-    const float* pre_t = precomp + ((int64_t)t * B + b) * G4;
-#pragma unroll
-    for (int k = 0; k < kMaxJ; ++k) {
-      if (k < nj) acc[k] = pre_t[tid + (k << 8)];
-    }
-    for (int h = 0; h < H; ++h) {
-      const float hv = s_h[h];
-      const float* wrow = w_hh_t + (int64_t)h * G4;
-#pragma unroll
-      for (int k = 0; k < kMaxJ; ++k) {
-        if (k < nj) acc[k] += hv * wrow[tid + (k << 8)];
+    // State update (+ prepare next step's masked state).
+    float* out_t = out + (int64_t)t * BH;
+    float* c_t = c_out + (int64_t)t * BH;
+    for (int64_t i = tid; i < BH; i += nthreads) {
+      const int b = i / H;
+      const int h = i - (int64_t)(i / H) * H;
+      const float* g4 = gates_t + (int64_t)b * G4;
+      const float gi = g4[h];
+      const float gf = g4[H + h];
+      const float gg = g4[2 * H + h];
+      const float go = g4[3 * H + h];
+      const float c_new = gf * cm_t[i] + gi * gg;
+      const float h_new = go * tanhf(c_new);
+      c_t[i] = c_new;
+      out_t[i] = h_new;
+      if (t + 1 < T) {
+        const float nd = notdone[(int64_t)(t + 1) * B + b];
+        hm[(int64_t)(t + 1) * BH + i] = nd * h_new;
+        cm[(int64_t)(t + 1) * BH + i] = nd * c_new;
+      } else {
+        hT[i] = h_new;
+        cT[i] = c_new;
       }
     }
-    float* gates_t = gates + ((int64_t)t * B + b) * (int64_t)G4;
-#pragma unroll
-    for (int k = 0; k < kMaxJ; ++k) {
-      if (k < nj) {
-        const int j = tid + (k << 8);
-        const float a = (j / H) == 2 ? tanhf(acc[k])
-                                     : 1.f / (1.f + __expf(-acc[k]));
-        s_g[j] = a;
-        gates_t[j] = a;
-      }
-    }
-    __syncthreads();
-
-    // State update.
-    for (int i = tid; i < H; i += blockDim.x) {
-      const float c_new = s_g[H + i] * s_c[i] + s_g[i] * s_g[2 * H + i];
-      const float h_new = s_g[3 * H + i] * tanhf(c_new);
-      s_c[i] = c_new;
-      s_h[i] = h_new;
-      out[(int64_t)t * BH + (int64_t)b * H + i] = h_new;
-      c_out[(int64_t)t * BH + (int64_t)b * H + i] = c_new;
-    }
-    __syncthreads();
-  }
-
-  for (int i = tid; i < H; i += blockDim.x) {
-    hT[(int64_t)b * H + i] = s_h[i];
-    cT[(int64_t)b * H + i] = s_c[i];
+    grid.sync();
   }
 }
 
-// Backward, one layer: same per-batch-element workgroup design (no grid
-// syncs). dh/dc carried in LDS; dgates staged in LDS for the dhm dot, whose
-// W_hh rows are read lane-coalesced.
+// Backward, one layer: computes pre-activation gate grads dgates [T,B,4H]
+// and the carried dh/dc. Phase 1 (elementwise over B*H) produces dgates;
+// phase 2 computes dh_{t-1} = dgates_t @ W_hh using a per-workgroup LDS
+// slice of W_hh^T held across all T steps (same persistent-weights design
+// as the forward). Weight/input grads are batched rocBLAS GEMMs on the host.
 __global__ __launch_bounds__(256) void lstm_bwd_kernel(
     const float* __restrict__ gates,    // [T,B,4H] post-activation
     const float* __restrict__ cm,       // [T,B,H]
     const float* __restrict__ c_out,    // [T,B,H]
-    const float* __restrict__ w_hh,     // [4H,H]
+    const float* __restrict__ w_hh_t,   // [H,4H] = W_hh^T (contiguous)
     const float* __restrict__ notdone,  // [T,B]
     const float* __restrict__ d_out,    // [T,B,H]
     const float* __restrict__ d_hT,     // [B,H]
     const float* __restrict__ d_cT,     // [B,H]
-    int T, int B, int H,
+    int T, int B, int H, int h_slice,
     float* __restrict__ dgates,  // [T,B,4H] PRE-activation grads
-    float* __restrict__ dh,      // [B,H] final d_h0
-    float* __restrict__ dc) {    // [B,H] final d_c0
-  extern __shared__ float smem[];
-  float* s_dh = smem;          // [H]
-  float* s_dc = s_dh + H;      // [H]
-  float* s_dg = s_dc + H;      // [4H]
-
-  const int b = blockIdx.x;
-  const int tid = threadIdx.x;
-  const int G4 = 4 * H;
+    float* __restrict__ dh,      // [B,H] workspace, ends as d_h0
+    float* __restrict__ dc) {    // [B,H] workspace, ends as d_c0
+  cg::grid_group grid = cg::this_grid();
+  extern __shared__ float s_wt[];  // [h_slice, 4H]
+  const int64_t tid = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  const int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
   const int64_t BH = (int64_t)B * H;
+  const int G4 = 4 * H;
 
-  for (int i = tid; i < H; i += blockDim.x) {
-    s_dh[i] = d_hT[(int64_t)b * H + i];
-    s_dc[i] = d_cT[(int64_t)b * H + i];
+  const int hh0 = blockIdx.x * h_slice;
+  const int hn = min(h_slice, H - hh0);
+
+  for (int i = threadIdx.x; i < hn * G4; i += blockDim.x) {
+    s_wt[i] = w_hh_t[(int64_t)hh0 * G4 + i];
   }
-  __syncthreads();
 
-  constexpr int kMaxI = 4;  // supports H up to 1024
-  const int ni = (H - tid + 255) / 256;
+  for (int64_t i = tid; i < BH; i += nthreads) {
+    dh[i] = d_hT[i];
+    dc[i] = d_cT[i];
+  }
+  grid.sync();
 
   for (int t = T - 1; t >= 0; --t) {
-    const float nd = notdone[(int64_t)t * B + b];
-    const float* gates_t = gates + ((int64_t)t * B + b) * (int64_t)G4;
-    float* dgates_t = dgates + ((int64_t)t * B + b) * (int64_t)G4;
+    const float* gates_t = gates + (int64_t)t * (int64_t)B * G4;
+    const float* cm_t = cm + (int64_t)t * BH;
+    const float* c_t = c_out + (int64_t)t * BH;
+    float* dgates_t = dgates + (int64_t)t * (int64_t)B * G4;
 
-    // Phase 1: gate gradients for this b; s_dc becomes masked d c_{t-1}.
-    for (int i = tid; i < H; i += blockDim.x) {
-      const float gi = gates_t[i];
-      const float gf = gates_t[H + i];
-      const float gg = gates_t[2 * H + i];
-      const float go = gates_t[3 * H + i];
+    // Phase 1: per-(b,h) gate gradients; dc becomes masked d c_{t-1}.
+    for (int64_t i = tid; i < BH; i += nthreads) {
+      const int b = i / H;
+      const int h = i - (int64_t)(i / H) * H;
+      const float* g4 = gates_t + (int64_t)b * G4;
+      const float gi = g4[h];
+      const float gf = g4[H + h];
+      const float gg = g4[2 * H + h];
+      const float go = g4[3 * H + h];
 
-      const float dh_t =
-          s_dh[i] + d_out[(int64_t)t * BH + (int64_t)b * H + i];
-      const float tc = tanhf(c_out[(int64_t)t * BH + (int64_t)b * H + i]);
-      const float dc_t = s_dc[i] + dh_t * go * (1.f - tc * tc);
+      const float dh_t = dh[i] + d_out[(int64_t)t * BH + i];
+      const float tc = tanhf(c_t[i]);
+      float dc_t = dc[i] + dh_t * go * (1.f - tc * tc);
 
       const float d_go = dh_t * tc;
       const float d_gi = dc_t * gg;
-      const float d_gf = dc_t * cm[(int64_t)t * BH + (int64_t)b * H + i];
+      const float d_gf = dc_t * cm_t[i];
       const float d_gg = dc_t * gi;
 
-      const float v_i = d_gi * gi * (1.f - gi);
-      const float v_f = d_gf * gf * (1.f - gf);
-      const float v_g = d_gg * (1.f - gg * gg);
-      const float v_o = d_go * go * (1.f - go);
-      s_dg[i] = v_i;
-      s_dg[H + i] = v_f;
-      s_dg[2 * H + i] = v_g;
-      s_dg[3 * H + i] = v_o;
-      dgates_t[i] = v_i;
-      dgates_t[H + i] = v_f;
-      dgates_t[2 * H + i] = v_g;
-      dgates_t[3 * H + i] = v_o;
+      float* dg4 = dgates_t + (int64_t)b * G4;
+      dg4[h] = d_gi * gi * (1.f - gi);
+      dg4[H + h] = d_gf * gf * (1.f - gf);
+      dg4[2 * H + h] = d_gg * (1.f - gg * gg);
+      dg4[3 * H + h] = d_go * go * (1.f - go);
 
-      s_dc[i] = nd * dc_t * gf;
+      const float nd = notdone[(int64_t)t * B + b];
+      dc[i] = nd * dc_t * gf;
     }
-    __syncthreads();
+    grid.sync();
 
-    // Phase 2: d h_{t-1}[h] = nd * sum_j s_dg[j] * W_hh[j, h].
-    float acc[kMaxI];
-#pragma unroll
-    for (int k = 0; k < kMaxI; ++k) acc[k] = 0.f;
-    for (int j = 0; j < G4; ++j) {
-      const float dgv = s_dg[j];
-      const float* wrow = w_hh + (int64_t)j * H;
-#pragma unroll
-      for (int k = 0; k < kMaxI; ++k) {
-        if (k < ni) acc[k] += dgv * wrow[tid + (k << 8)];
-      }
+    // Phase 2: dh_{t-1}[b, h] for this WG's h-slice, from the LDS W_hh^T.
+    for (int i = threadIdx.x; i < hn * B; i += blockDim.x) {
+      const int b = i / hn;
+      const int hl = i - (i / hn) * hn;
+      const float* dg4 = dgates_t + (int64_t)b * G4;
+      const float* wrow = s_wt + (int64_t)hl * G4;
+      float acc = 0.f;
+      for (int j = 0; j < G4; ++j) acc += dg4[j] * wrow[j];
+      const float nd = notdone[(int64_t)t * B + b];
+      dh[(int64_t)b * H + hh0 + hl] = nd * acc;
     }
-    __syncthreads();  // all reads of s_dh done before overwrite
-#pragma unroll
-    for (int k = 0; k < kMaxI; ++k) {
-      if (k < ni) s_dh[tid + (k << 8)] = nd * acc[k];
-    }
-    __syncthreads();
-  }
-
-  for (int i = tid; i < H; i += blockDim.x) {
-    dh[(int64_t)b * H + i] = s_dh[i];
-    dc[(int64_t)b * H + i] = s_dc[i];
+    grid.sync();
   }
 }
 
@@ -609,13 +590,21 @@ std::vector<torch::Tensor> lstm_unroll_fwd(
   torch::Tensor out;
 
   const int threads = 256;
-  TORCH_CHECK(H <= 1024, "lstm hidden size > 1024 not supported");
-  const size_t fwd_lds = (size_t)6 * H * sizeof(float);
+  // Workgroup count: each WG owns a W_hh row-slice held in LDS. Cap the
+  // per-WG slice at ~48 KiB so several blocks fit per CU.
+  const int64_t w_bytes = (int64_t)4 * H * H * 4;
+  int grid = std::max<int64_t>(128, ceil_div(w_bytes, 48 * 1024));
+  const int j_slice = ceil_div((int64_t)4 * H, grid);
+  const size_t fwd_lds = (size_t)j_slice * H * sizeof(float);
+  TORCH_CHECK(fwd_lds <= 64 * 1024, "lstm hidden size too large for LDS slices");
+  {
+    int cap = coop_grid_size((const void*)lstm_fwd_kernel, threads, fwd_lds);
+    TORCH_CHECK(grid <= cap, "lstm fwd grid exceeds cooperative capacity");
+  }
 
   for (int l = 0; l < L; ++l) {
     auto w_ih = flat_weights[4 * l].contiguous();
     auto w_hh = flat_weights[4 * l + 1].contiguous();
-    auto w_hh_t = w_hh.t().contiguous();  // [H, 4H], lane-coalesced reads
     auto bias = (flat_weights[4 * l + 2] + flat_weights[4 * l + 3]);
 
     // The parallel 90%: one GEMM over all T*B rows.
@@ -633,15 +622,26 @@ std::vector<torch::Tensor> lstm_unroll_fwd(
     auto h0_l = h0[l].contiguous();
     auto c0_l = c0[l].contiguous();
 
-    hipLaunchKernelGGL(lstm_fwd_kernel, dim3(B), dim3(threads), fwd_lds,
-                       stream, precomp.data_ptr<float>(),
-                       w_hh_t.data_ptr<float>(),
-                       notdone_f.data_ptr<float>(), h0_l.data_ptr<float>(),
-                       c0_l.data_ptr<float>(), T, B, H,
-                       out.data_ptr<float>(), gates.data_ptr<float>(),
-                       hm.data_ptr<float>(), cm.data_ptr<float>(),
-                       c_out.data_ptr<float>(), hT[l].data_ptr<float>(),
-                       cT[l].data_ptr<float>());
+    const float* precomp_p = precomp.data_ptr<float>();
+    const float* w_hh_p = w_hh.data_ptr<float>();
+    const float* notdone_p = notdone_f.data_ptr<float>();
+    const float* h0_p = h0_l.data_ptr<float>();
+    const float* c0_p = c0_l.data_ptr<float>();
+    float* out_p = out.data_ptr<float>();
+    float* gates_p = gates.data_ptr<float>();
+    float* hm_p = hm.data_ptr<float>();
+    float* cm_p = cm.data_ptr<float>();
+    float* c_out_p = c_out.data_ptr<float>();
+    float* hT_p = hT[l].data_ptr<float>();
+    float* cT_p = cT[l].data_ptr<float>();
+    int T_ = T, B_ = B, H_ = H, j_slice_ = j_slice;
+    void* args[] = {&precomp_p, &w_hh_p, &notdone_p, &h0_p, &c0_p,
+                    &T_,        &B_,     &H_,        &j_slice_,
+                    &out_p,     &gates_p, &hm_p,     &cm_p,
+                    &c_out_p,   &hT_p,   &cT_p};
+    DEVCHECK(hipLaunchCooperativeKernel((const void*)lstm_fwd_kernel,
+                                        dim3(grid), dim3(threads), args,
+                                        fwd_lds, stream));
 
     stash.push_back(layer_in);
     stash.push_back(gates);
@@ -674,8 +674,15 @@ std::vector<torch::Tensor> lstm_unroll_bwd(
   std::vector<torch::Tensor> d_weights(4 * L);
 
   const int threads = 256;
-  TORCH_CHECK(H <= 1024, "lstm hidden size > 1024 not supported");
-  const size_t bwd_lds = (size_t)6 * H * sizeof(float);
+  const int64_t w_bytes = (int64_t)4 * H * H * 4;
+  int grid = std::max<int64_t>(128, ceil_div(w_bytes, 48 * 1024));
+  const int h_slice = ceil_div((int64_t)H, grid);
+  const size_t bwd_lds = (size_t)h_slice * 4 * H * sizeof(float);
+  TORCH_CHECK(bwd_lds <= 64 * 1024, "lstm hidden size too large for LDS slices");
+  {
+    int cap = coop_grid_size((const void*)lstm_bwd_kernel, threads, bwd_lds);
+    TORCH_CHECK(grid <= cap, "lstm bwd grid exceeds cooperative capacity");
+  }
 
   torch::Tensor d_out = d_out_top.contiguous();
   for (int l = L - 1; l >= 0; --l) {
@@ -686,6 +693,7 @@ std::vector<torch::Tensor> lstm_unroll_bwd(
     auto c_out = stash[5 * l + 4];
     auto w_ih = flat_weights[4 * l].contiguous();
     auto w_hh = flat_weights[4 * l + 1].contiguous();
+    auto w_hh_t = w_hh.t().contiguous();  // [H, 4H] for the bwd LDS slices
 
     auto dgates = torch::empty({T, B, 4 * H}, opts);
     auto dh = torch::empty({B, H}, opts);
@@ -693,13 +701,24 @@ std::vector<torch::Tensor> lstm_unroll_bwd(
     auto d_hT_l = d_hT[l].contiguous();
     auto d_cT_l = d_cT[l].contiguous();
 
-    hipLaunchKernelGGL(lstm_bwd_kernel, dim3(B), dim3(threads), bwd_lds,
-                       stream, gates.data_ptr<float>(), cm.data_ptr<float>(),
-                       c_out.data_ptr<float>(), w_hh.data_ptr<float>(),
-                       notdone_f.data_ptr<float>(), d_out.data_ptr<float>(),
-                       d_hT_l.data_ptr<float>(), d_cT_l.data_ptr<float>(),
-                       T, B, H, dgates.data_ptr<float>(),
-                       dh.data_ptr<float>(), dc.data_ptr<float>());
+    const float* gates_p = gates.data_ptr<float>();
+    const float* cm_p = cm.data_ptr<float>();
+    const float* c_out_p = c_out.data_ptr<float>();
+    const float* w_hh_t_p = w_hh_t.data_ptr<float>();
+    const float* notdone_p = notdone_f.data_ptr<float>();
+    const float* d_out_p = d_out.data_ptr<float>();
+    const float* d_hT_p = d_hT_l.data_ptr<float>();
+    const float* d_cT_p = d_cT_l.data_ptr<float>();
+    float* dgates_p = dgates.data_ptr<float>();
+    float* dh_p = dh.data_ptr<float>();
+    float* dc_p = dc.data_ptr<float>();
+    int T_ = T, B_ = B, H_ = H, h_slice_ = h_slice;
+    void* args[] = {&gates_p, &cm_p, &c_out_p, &w_hh_t_p, &notdone_p,
+                    &d_out_p, &d_hT_p, &d_cT_p, &T_, &B_, &H_, &h_slice_,
+                    &dgates_p, &dh_p, &dc_p};
+    DEVCHECK(hipLaunchCooperativeKernel((const void*)lstm_bwd_kernel,
+                                        dim3(grid), dim3(threads), args,
+                                        bwd_lds, stream));
 
     // Weight/input grads: plain GEMMs (rocBLAS).
     auto dg2 = dgates.reshape({(int64_t)T * B, 4 * H});
