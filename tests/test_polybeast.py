@@ -90,3 +90,23 @@ def test_train_against_socket_env_servers(tmp_path):
     finally:
         for s in servers:
             s.stop()
+
+
+def test_combined_cli_launcher(tmp_path):
+    """polybeast.py combined CLI (ref: torchbeast/polybeast.py) end-to-end
+    as a subprocess with in-process synthetic envs."""
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "-m", "torchbeast_amd.polybeast",
+         "--env", "synthetic:4x36x36:6", "--num_actors", "2",
+         "--batch_size", "2", "--unroll_length", "8", "--total_steps", "32",
+         "--num_learner_threads", "1", "--num_inference_threads", "1",
+         "--disable_cuda", "--savedir", str(tmp_path), "--xpid", "combined"],
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        timeout=240,
+        capture_output=True,
+    )
+    assert out.returncode == 0, out.stderr.decode()[-2000:]
+    assert os.path.exists(os.path.join(str(tmp_path), "combined", "model.tar"))

@@ -8,6 +8,8 @@ Build with `python setup.py build_ext --inplace` (the extension is built
 in-tree so it travels with the repo snapshot).
 """
 
+import torch  # noqa: F401  (loads libc10/libtorch before the extension)
+
 try:
     from torchbeast_amd.runtime._tbruntime import (  # noqa: F401
         ActorPool,
