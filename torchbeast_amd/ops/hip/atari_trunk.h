@@ -31,4 +31,9 @@ std::vector<torch::Tensor> atari_trunk_bwd(
     torch::Tensor w3, torch::Tensor out1, torch::Tensor out2,
     torch::Tensor out3_flat, torch::Tensor d_out3);
 
+// bf16 MFMA GEMM (mfma_gemm.hip): C[M,N] f32 = A[M,K] @ B[N,K]^T.
+torch::Tensor mfma_gemm(torch::Tensor A, torch::Tensor B);
+torch::Tensor mfma_gemm_probe(torch::Tensor A, torch::Tensor B,
+                              int64_t variant);
+
 }  // namespace tbamd
