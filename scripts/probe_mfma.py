@@ -1,3 +1,5 @@
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch, timeit
 from torchbeast_amd.ops import _tbops
 torch.manual_seed(0)
