@@ -1,7 +1,8 @@
 import os, sys
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch, timeit
-from torchbeast_amd.ops import _tbops
+import torchbeast_amd.ops as _ops
+_tbops = _ops.require_ext()
 torch.manual_seed(0)
 M,N,K = 128, 128, 64
 A = torch.randn(M,K,device="cuda").bfloat16()
