@@ -35,5 +35,6 @@ std::vector<torch::Tensor> atari_trunk_bwd(
 torch::Tensor mfma_gemm(torch::Tensor A, torch::Tensor B);
 torch::Tensor mfma_gemm_probe(torch::Tensor A, torch::Tensor B,
                               int64_t variant);
+torch::Tensor mfma_gemm_v2(torch::Tensor A, torch::Tensor B);
 
 }  // namespace tbamd

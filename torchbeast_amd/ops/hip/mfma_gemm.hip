@@ -156,7 +156,121 @@ torch::Tensor launch(torch::Tensor A, torch::Tensor B) {
   return C;
 }
 
+// ---------------------------------------------------------------------------
+// v2: 128x128 tile, 4 waves (64x64 sub-tile each, 4x4 fragments), direct
+// global->LDS staging via __builtin_amdgcn_global_load_lds (width 16). The
+// LDS destination of global_load_lds is wave-uniform-base + lane*16, so the
+// [128][32] bf16 tiles stay linear row-major and each wave stages a 32-row
+// band (lane -> row = lane/4, 16B chunk = lane%4).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void mfma_gemm_v2_kernel(
+    const __bf16* __restrict__ A,  // [M, K]
+    const __bf16* __restrict__ B,  // [N, K]
+    float* __restrict__ C,         // [M, N]
+    int M, int N, int K) {
+  constexpr int BM = 128, BN = 128, BK = 32;
+  __shared__ __bf16 sA[BM][BK];
+  __shared__ __bf16 sB[BN][BK];
+
+  const int tile_m = blockIdx.x * BM;
+  const int tile_n = blockIdx.y * BN;
+  const int tid = threadIdx.x;
+  const int wave = tid / 64;
+  const int lane = tid % 64;
+  const int wm = (wave >> 1) * 64;  // wave's 64x64 sub-tile origin
+  const int wn = (wave & 1) * 64;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int a = 0; a < 4; ++a)
+#pragma unroll
+    for (int b = 0; b < 4; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+
+  // Staging geometry: each wave loads a 32-row band of each tile; a band is
+  // 32 rows x 64 B = 2 KiB = two 1 KiB (64 lanes x 16 B) issues.
+  const int band = wave * 32;
+  const int lrow = lane / 4;          // 0..15
+  const int lchunk = (lane % 4) * 8;  // bf16 elements: 0, 8, 16, 24
+
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      // Hardware semantics: lane l's 16 B land at lds_base + l*16 (the
+      // base is wave-uniform). With row = lane/4 and chunk = lane%4 the
+      // linear row-major tile offset equals lane*16 exactly, so the base
+      // is the start of this 16-row half-band.
+      const int row = band + half * 16 + lrow;
+      typedef const __attribute__((address_space(1))) uint32_t* gptr_t;
+      typedef __attribute__((address_space(3))) uint32_t* lptr_t;
+      __builtin_amdgcn_global_load_lds(
+          (gptr_t)(&A[(int64_t)(tile_m + row) * K + k0 + lchunk]),
+          (lptr_t)(&sA[band + half * 16][0]), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (gptr_t)(&B[(int64_t)(tile_n + row) * K + k0 + lchunk]),
+          (lptr_t)(&sB[band + half * 16][0]), 16, 0, 0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    // 16 MFMA per wave per K-step: 4x4 fragments of 16x16x32.
+    bf16x8 af[4], bf[4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      af[m] = *reinterpret_cast<const bf16x8*>(
+          &sA[wm + m * 16 + (lane & 15)][8 * (lane >> 4)]);
+    }
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      bf[n] = *reinterpret_cast<const bf16x8*>(
+          &sB[wn + n * 16 + (lane & 15)][8 * (lane >> 4)]);
+    }
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[m], bf[n], acc[m][n], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = tile_m + wm + m * 16 + (lane >> 4) * 4 + r;
+        const int col = tile_n + wn + n * 16 + (lane & 15);
+        C[(int64_t)row * N + col] = acc[m][n][r];
+      }
+    }
+  }
+}
+
 }  // namespace
+
+torch::Tensor mfma_gemm_v2(torch::Tensor A, torch::Tensor B) {
+  const int M = A.size(0), K = A.size(1), N = B.size(0);
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16 &&
+                  B.scalar_type() == torch::kBFloat16,
+              "mfma_gemm_v2: bf16 GPU tensors required");
+  TORCH_CHECK(B.size(1) == K, "mfma_gemm_v2: K mismatch");
+  TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 32 == 0,
+              "mfma_gemm_v2: M,N multiple of 128, K multiple of 32");
+  auto C = torch::empty({M, N}, A.options().dtype(torch::kFloat32));
+  auto Ac = A.contiguous();
+  auto Bc = B.contiguous();
+  hipLaunchKernelGGL(mfma_gemm_v2_kernel, dim3(M / 128, N / 128), dim3(256),
+                     0, at::cuda::getCurrentCUDAStream(),
+                     reinterpret_cast<const __bf16*>(Ac.data_ptr()),
+                     reinterpret_cast<const __bf16*>(Bc.data_ptr()),
+                     C.data_ptr<float>(), M, N, K);
+  return C;
+}
 
 // variant: 0 = (layout0, noswap), 1 = (layout1, noswap),
 //          2 = (layout0, swap),   3 = (layout1, swap).

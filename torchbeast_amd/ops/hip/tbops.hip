@@ -750,6 +750,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("atari_trunk_supported", &tbamd::atari_trunk_supported);
   m.def("mfma_gemm", &tbamd::mfma_gemm);
   m.def("mfma_gemm_probe", &tbamd::mfma_gemm_probe);
+  m.def("mfma_gemm_v2", &tbamd::mfma_gemm_v2);
   m.def("vtrace_from_logits", &vtrace_from_logits);
   m.def("fused_impala_loss_fwd", &fused_impala_loss_fwd);
   m.def("rmsprop_step", &rmsprop_step);
