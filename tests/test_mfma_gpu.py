@@ -23,6 +23,17 @@ def test_mfma_gemm_matches_matmul(M, N, K):
     torch.testing.assert_close(C, ref, rtol=1e-2, atol=1e-2)
 
 
+@pytest.mark.parametrize("M,N,K", [(128, 128, 64), (256, 384, 512)])
+def test_mfma_gemm_v2_matches_matmul(M, N, K):
+    ext = ops.require_ext()
+    torch.manual_seed(1)
+    A = torch.randn(M, K, device="cuda").bfloat16()
+    B = torch.randn(N, K, device="cuda").bfloat16()
+    C = ext.mfma_gemm_v2(A, B)
+    ref = A.float() @ B.float().t()
+    torch.testing.assert_close(C, ref, rtol=1e-2, atol=1e-2)
+
+
 def test_mfma_gemm_rejects_bad_shapes():
     ext = ops.require_ext()
     A = torch.randn(65, 32, device="cuda").bfloat16()
