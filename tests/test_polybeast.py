@@ -110,3 +110,11 @@ def test_combined_cli_launcher(tmp_path):
     )
     assert out.returncode == 0, out.stderr.decode()[-2000:]
     assert os.path.exists(os.path.join(str(tmp_path), "combined", "model.tar"))
+
+
+def test_polybeast_test_mode(tmp_path):
+    """test mode loads the checkpoint and runs greedy local-env episodes."""
+    flags = _flags(tmp_path, total_steps=32)
+    polybeast_learner.train(flags)
+    avg = polybeast_learner.test(flags, num_episodes=1)
+    assert isinstance(avg, float)

@@ -529,13 +529,44 @@ def train(flags):  # noqa: C901
     logging.info("Rank %d done after %d updates.", rank, update_counter["done"])
 
 
-def test(flags):
-    """Greedy-policy evaluation. The reference leaves polybeast test
-    unimplemented (ref: polybeast_learner.py:596-597); here we delegate to
-    the monobeast-style local-env evaluator."""
+def test(flags, num_episodes: int = 10):
+    """Greedy-policy evaluation on a local env (the reference leaves
+    polybeast test unimplemented, ref: polybeast_learner.py:596-597)."""
     from torchbeast_amd import monobeast
+    from torchbeast_amd.core import environment
 
-    return monobeast.test(flags)
+    if flags.xpid is None:
+        checkpointpath = "./latest/model.tar"
+    else:
+        checkpointpath = os.path.expandvars(
+            os.path.expanduser(f"{flags.savedir}/{flags.xpid}/model.tar")
+        )
+
+    model = create_model(flags)
+    model.eval()
+    checkpoint = torch.load(checkpointpath, map_location="cpu",
+                            weights_only=False)
+    model.load_state_dict(checkpoint["model_state_dict"])
+
+    env = environment.Environment(monobeast.create_env(flags))
+    observation = env.initial()
+    core_state = model.initial_state(batch_size=1)
+    returns = []
+    while len(returns) < num_episodes:
+        outputs, core_state = model(observation, core_state)
+        action = _as_agent_output(outputs)[0]
+        observation = env.step(action)
+        if observation["done"].item():
+            returns.append(observation["episode_return"].item())
+            logging.info(
+                "Episode ended after %d steps. Return: %.1f",
+                observation["episode_step"].item(),
+                observation["episode_return"].item(),
+            )
+    env.close()
+    logging.info("Average returns over %i episodes: %.1f", num_episodes,
+                 sum(returns) / len(returns))
+    return sum(returns) / len(returns)
 
 
 def main(flags):
