@@ -50,3 +50,22 @@ def test_graft_smoke():
     import __graft_entry__
 
     __graft_entry__.smoke()
+
+
+def test_monobeast_train_short_gpu(tmp_path):
+    """Monobeast on GPU: CPU actor processes + shared-memory buffers feed a
+    GPU learner running the fused HIP ops."""
+    from torchbeast_amd import monobeast
+
+    flags = monobeast.parser.parse_args([])
+    flags.env = "synthetic:4x84x84:6"
+    flags.savedir = str(tmp_path)
+    flags.xpid = "monogpu"
+    flags.num_actors = 2
+    flags.num_buffers = 6
+    flags.batch_size = 2
+    flags.unroll_length = 16
+    flags.total_steps = 16 * 2 * 5
+    flags.num_learner_threads = 1
+    monobeast.train(flags)
+    assert os.path.exists(os.path.join(str(tmp_path), "monogpu", "model.tar"))
