@@ -98,9 +98,10 @@ class InferenceRunner {
       try {
         serve(*batch, stream);
       } catch (const std::exception& e) {
-        // Drop the batch: promises break with AsyncError; actors surface it.
-        fprintf(stderr, "InferenceRunner error: %s\n", e.what());
-        return;
+        // Drop this batch (its promises break with AsyncError and the
+        // affected actors surface it) but keep the engine serving.
+        fprintf(stderr, "InferenceRunner error (batch dropped): %s\n",
+                e.what());
       }
     }
   }
