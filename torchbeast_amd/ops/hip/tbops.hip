@@ -374,18 +374,18 @@ torch::Tensor policy_sample(torch::Tensor logits, bool greedy, int64_t seed) {
 // ---------------------------------------------------------------------------
 
 // Forward, one layer, persistent cooperative kernel. Each workgroup OWNS a
-// slice of the HIDDEN dimension: it stages the W_hh rows of all four gates
-// for its h-slice in LDS (held across the whole unroll) and computes both
-// the gate dots AND the state update for those rows — so each step needs
-// only ONE grid-wide sync (publishing hm[t+1]; grid.sync costs ~100 us
-// across MI355X's 8 XCDs, making sync count the dominant term).
+// slice of W_hh rows staged in LDS for the whole unroll (one load instead of
+// T re-reads from L3 — W_hh is ~4 MB for hidden 519, larger than one XCD's
+// L2). Per step: gate phase (each WG computes its gate slice for all B) ->
+// grid.sync -> state-update phase (grid-stride over B*H, which also writes
+// the next step's done-masked state) -> grid.sync.
 __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     const float* __restrict__ precomp,  // [T,B,4H] = x@W_ih^T + b_ih + b_hh
     const float* __restrict__ w_hh,     // [4H,H]
     const float* __restrict__ notdone,  // [T,B]
     const float* __restrict__ h0,       // [B,H]
     const float* __restrict__ c0,       // [B,H]
-    int T, int B, int H, int h_slice,
+    int T, int B, int H, int j_slice,
     float* __restrict__ out,    // [T,B,H]
     float* __restrict__ gates,  // [T,B,4H] post-activation
     float* __restrict__ hm,     // [T,B,H] masked h_{t-1}
@@ -394,26 +394,21 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     float* __restrict__ hT,     // [B,H]
     float* __restrict__ cT) {   // [B,H]
   cg::grid_group grid = cg::this_grid();
-  extern __shared__ float s_w[];  // [4*h_slice][H+1] (padded rows)
+  extern __shared__ float s_w[];  // [j_slice, H]
   const int64_t tid = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   const int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
   const int64_t BH = (int64_t)B * H;
   const int G4 = 4 * H;
-  const int HP = H + 1;
 
-  const int h0i = blockIdx.x * h_slice;
-  const int hn = min(h_slice, H - h0i);
+  const int j0 = blockIdx.x * j_slice;
+  const int jn = min(j_slice, G4 - j0);
 
-  // Stage W_hh rows {g*H + h : g in 0..3, h in slice} once for the unroll.
-  for (int i = threadIdx.x; i < 4 * hn * H; i += blockDim.x) {
-    const int row = i / H;  // g*hn + hl
-    const int hp = i - row * H;
-    const int g = row / hn;
-    const int hl = row - g * hn;
-    s_w[row * HP + hp] = w_hh[((int64_t)g * H + h0i + hl) * H + hp];
+  // Stage this workgroup's W_hh rows into LDS (once for the whole unroll).
+  for (int i = threadIdx.x; i < jn * H; i += blockDim.x) {
+    s_w[i] = w_hh[(int64_t)j0 * H + i];
   }
 
-  // Initialize hm/cm for t=0 (grid-stride; published by the sync below).
+  // Initialize hm/cm for t=0.
   for (int64_t i = tid; i < BH; i += nthreads) {
     const int b = i / H;
     const float nd = notdone[b];
@@ -427,60 +422,58 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     const float* cm_t = cm + (int64_t)t * BH;
     float* gates_t = gates + (int64_t)t * (int64_t)B * G4;
     const float* pre_t = precomp + (int64_t)t * (int64_t)B * G4;
+
+    // Gate phase: this WG computes gates[b, j0..j0+jn) for all b.
+    for (int i = threadIdx.x; i < jn * B; i += blockDim.x) {
+      const int b = i / jn;           // lanes iterate j fastest: the hm row
+      const int jl = i - (i / jn) * jn;  // is shared across the wave.
+      const float* hrow = hm_t + (int64_t)b * H;
+      const float* wrow = s_w + (int64_t)jl * H;
+      float acc = pre_t[(int64_t)b * G4 + j0 + jl];
+      for (int h = 0; h < H; ++h) acc += hrow[h] * wrow[h];
+      const int gate = (j0 + jl) / H;
+      if (gate == 2) {
+        acc = tanhf(acc);
+      } else {
+        acc = 1.f / (1.f + __expf(-acc));
+      }
+      gates_t[(int64_t)b * G4 + j0 + jl] = acc;
+    }
+    grid.sync();
+
+    // State update (+ prepare next step's masked state).
     float* out_t = out + (int64_t)t * BH;
     float* c_t = c_out + (int64_t)t * BH;
-
-    for (int i = threadIdx.x; i < B * hn; i += blockDim.x) {
-      const int b = i / hn;
-      const int hl = i - (i / hn) * hn;
-      const int h = h0i + hl;
-      const float* hrow = hm_t + (int64_t)b * H;
-      const float* w0 = s_w + (0 * hn + hl) * HP;
-      const float* w1 = s_w + (1 * hn + hl) * HP;
-      const float* w2 = s_w + (2 * hn + hl) * HP;
-      const float* w3 = s_w + (3 * hn + hl) * HP;
-      float a0 = pre_t[(int64_t)b * G4 + h];
-      float a1 = pre_t[(int64_t)b * G4 + H + h];
-      float a2 = pre_t[(int64_t)b * G4 + 2 * H + h];
-      float a3 = pre_t[(int64_t)b * G4 + 3 * H + h];
-      for (int hp = 0; hp < H; ++hp) {
-        const float v = hrow[hp];
-        a0 += v * w0[hp];
-        a1 += v * w1[hp];
-        a2 += v * w2[hp];
-        a3 += v * w3[hp];
-      }
-      const float gi = 1.f / (1.f + __expf(-a0));
-      const float gf = 1.f / (1.f + __expf(-a1));
-      const float gg = tanhf(a2);
-      const float go = 1.f / (1.f + __expf(-a3));
-      float* g4 = gates_t + (int64_t)b * G4;
-      g4[h] = gi;
-      g4[H + h] = gf;
-      g4[2 * H + h] = gg;
-      g4[3 * H + h] = go;
-
-      const int64_t bi = (int64_t)b * H + h;
-      const float c_new = gf * cm_t[bi] + gi * gg;
+    for (int64_t i = tid; i < BH; i += nthreads) {
+      const int b = i / H;
+      const int h = i - (int64_t)(i / H) * H;
+      const float* g4 = gates_t + (int64_t)b * G4;
+      const float gi = g4[h];
+      const float gf = g4[H + h];
+      const float gg = g4[2 * H + h];
+      const float go = g4[3 * H + h];
+      const float c_new = gf * cm_t[i] + gi * gg;
       const float h_new = go * tanhf(c_new);
-      c_t[bi] = c_new;
-      out_t[bi] = h_new;
+      c_t[i] = c_new;
+      out_t[i] = h_new;
       if (t + 1 < T) {
         const float nd = notdone[(int64_t)(t + 1) * B + b];
-        hm[(int64_t)(t + 1) * BH + bi] = nd * h_new;
-        cm[(int64_t)(t + 1) * BH + bi] = nd * c_new;
+        hm[(int64_t)(t + 1) * BH + i] = nd * h_new;
+        cm[(int64_t)(t + 1) * BH + i] = nd * c_new;
       } else {
-        hT[bi] = h_new;
-        cT[bi] = c_new;
+        hT[i] = h_new;
+        cT[i] = c_new;
       }
     }
     grid.sync();
   }
 }
 
-// Backward, one layer: same h-slice ownership as the forward — each WG
-// computes gate grads AND the dhm dot for its hidden rows, so one
-// grid.sync per step (publishing dgates) plus an intra-WG barrier.
+// Backward, one layer: computes pre-activation gate grads dgates [T,B,4H]
+// and the carried dh/dc. Phase 1 (elementwise over B*H) produces dgates;
+// phase 2 computes dh_{t-1} = dgates_t @ W_hh using a per-workgroup LDS
+// slice of W_hh^T held across all T steps (same persistent-weights design
+// as the forward). Weight/input grads are batched rocBLAS GEMMs on the host.
 __global__ __launch_bounds__(256) void lstm_bwd_kernel(
     const float* __restrict__ gates,    // [T,B,4H] post-activation
     const float* __restrict__ cm,       // [T,B,H]
@@ -495,20 +488,17 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
     float* __restrict__ dh,      // [B,H] workspace, ends as d_h0
     float* __restrict__ dc) {    // [B,H] workspace, ends as d_c0
   cg::grid_group grid = cg::this_grid();
-  extern __shared__ float s_wt[];  // [h_slice][4H+1] (padded rows)
+  extern __shared__ float s_wt[];  // [h_slice, 4H]
   const int64_t tid = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   const int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
   const int64_t BH = (int64_t)B * H;
   const int G4 = 4 * H;
-  const int GP = G4 + 1;
 
   const int hh0 = blockIdx.x * h_slice;
   const int hn = min(h_slice, H - hh0);
 
   for (int i = threadIdx.x; i < hn * G4; i += blockDim.x) {
-    const int hl = i / G4;
-    const int j = i - hl * G4;
-    s_wt[hl * GP + j] = w_hh_t[(int64_t)(hh0 + hl) * G4 + j];
+    s_wt[i] = w_hh_t[(int64_t)hh0 * G4 + i];
   }
 
   for (int64_t i = tid; i < BH; i += nthreads) {
@@ -523,25 +513,23 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
     const float* c_t = c_out + (int64_t)t * BH;
     float* dgates_t = dgates + (int64_t)t * (int64_t)B * G4;
 
-    // Phase 1: gate gradients for this WG's h-slice; dc -> masked d c_{t-1}.
-    for (int i = threadIdx.x; i < B * hn; i += blockDim.x) {
-      const int b = i / hn;
-      const int hl = i - (i / hn) * hn;
-      const int h = hh0 + hl;
-      const int64_t bi = (int64_t)b * H + h;
+    // Phase 1: per-(b,h) gate gradients; dc becomes masked d c_{t-1}.
+    for (int64_t i = tid; i < BH; i += nthreads) {
+      const int b = i / H;
+      const int h = i - (int64_t)(i / H) * H;
       const float* g4 = gates_t + (int64_t)b * G4;
       const float gi = g4[h];
       const float gf = g4[H + h];
       const float gg = g4[2 * H + h];
       const float go = g4[3 * H + h];
 
-      const float dh_t = dh[bi] + d_out[(int64_t)t * BH + bi];
-      const float tc = tanhf(c_t[bi]);
-      const float dc_t = dc[bi] + dh_t * go * (1.f - tc * tc);
+      const float dh_t = dh[i] + d_out[(int64_t)t * BH + i];
+      const float tc = tanhf(c_t[i]);
+      float dc_t = dc[i] + dh_t * go * (1.f - tc * tc);
 
       const float d_go = dh_t * tc;
       const float d_gi = dc_t * gg;
-      const float d_gf = dc_t * cm_t[bi];
+      const float d_gf = dc_t * cm_t[i];
       const float d_gg = dc_t * gi;
 
       float* dg4 = dgates_t + (int64_t)b * G4;
@@ -551,23 +539,22 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
       dg4[3 * H + h] = d_go * go * (1.f - go);
 
       const float nd = notdone[(int64_t)t * B + b];
-      dc[bi] = nd * dc_t * gf;
+      dc[i] = nd * dc_t * gf;
     }
-    grid.sync();  // publish dgates[t] for the full-j dot below
+    grid.sync();
 
-    // Phase 2: dh_{t-1}[b, h] for this slice from the LDS W_hh^T.
+    // Phase 2: dh_{t-1}[b, h] for this WG's h-slice, from the LDS W_hh^T.
     for (int i = threadIdx.x; i < hn * B; i += blockDim.x) {
       const int b = i / hn;
       const int hl = i - (i / hn) * hn;
       const float* dg4 = dgates_t + (int64_t)b * G4;
-      const float* wrow = s_wt + (int64_t)hl * GP;
+      const float* wrow = s_wt + (int64_t)hl * G4;
       float acc = 0.f;
       for (int j = 0; j < G4; ++j) acc += dg4[j] * wrow[j];
       const float nd = notdone[(int64_t)t * B + b];
       dh[(int64_t)b * H + hh0 + hl] = nd * acc;
     }
-    // Next iteration's phase 1 reads dh/dc of this slice only: intra-WG.
-    __syncthreads();
+    grid.sync();
   }
 }
 
@@ -603,12 +590,13 @@ std::vector<torch::Tensor> lstm_unroll_fwd(
   torch::Tensor out;
 
   const int threads = 256;
-  // h-slice sizing: LDS holds 4*h_slice padded W_hh rows; target <=140 KiB
-  // and at least ~16 workgroups of parallelism.
-  int h_slice = std::max<int64_t>(1, (140 * 1024 / 4) / (4 * (H + 1)));
-  h_slice = std::min<int64_t>(h_slice, std::max<int64_t>(1, ceil_div(H, 16)));
-  const int grid = ceil_div(H, h_slice);
-  const size_t fwd_lds = (size_t)4 * h_slice * (H + 1) * sizeof(float);
+  // Workgroup count: each WG owns a W_hh row-slice held in LDS. Cap the
+  // per-WG slice at ~48 KiB so several blocks fit per CU.
+  const int64_t w_bytes = (int64_t)4 * H * H * 4;
+  int grid = std::max<int64_t>(128, ceil_div(w_bytes, 48 * 1024));
+  const int j_slice = ceil_div((int64_t)4 * H, grid);
+  const size_t fwd_lds = (size_t)j_slice * H * sizeof(float);
+  TORCH_CHECK(fwd_lds <= 64 * 1024, "lstm hidden size too large for LDS slices");
   {
     int cap = coop_grid_size((const void*)lstm_fwd_kernel, threads, fwd_lds);
     TORCH_CHECK(grid <= cap, "lstm fwd grid exceeds cooperative capacity");
@@ -646,9 +634,9 @@ std::vector<torch::Tensor> lstm_unroll_fwd(
     float* c_out_p = c_out.data_ptr<float>();
     float* hT_p = hT[l].data_ptr<float>();
     float* cT_p = cT[l].data_ptr<float>();
-    int T_ = T, B_ = B, H_ = H, h_slice_ = h_slice;
+    int T_ = T, B_ = B, H_ = H, j_slice_ = j_slice;
     void* args[] = {&precomp_p, &w_hh_p, &notdone_p, &h0_p, &c0_p,
-                    &T_,        &B_,     &H_,        &h_slice_,
+                    &T_,        &B_,     &H_,        &j_slice_,
                     &out_p,     &gates_p, &hm_p,     &cm_p,
                     &c_out_p,   &hT_p,   &cT_p};
     DEVCHECK(hipLaunchCooperativeKernel((const void*)lstm_fwd_kernel,
@@ -686,10 +674,11 @@ std::vector<torch::Tensor> lstm_unroll_bwd(
   std::vector<torch::Tensor> d_weights(4 * L);
 
   const int threads = 256;
-  int h_slice = std::max<int64_t>(1, (140 * 1024 / 4) / (4 * H + 1));
-  h_slice = std::min<int64_t>(h_slice, std::max<int64_t>(1, ceil_div(H, 16)));
-  const int grid = ceil_div(H, h_slice);
-  const size_t bwd_lds = (size_t)h_slice * (4 * H + 1) * sizeof(float);
+  const int64_t w_bytes = (int64_t)4 * H * H * 4;
+  int grid = std::max<int64_t>(128, ceil_div(w_bytes, 48 * 1024));
+  const int h_slice = ceil_div((int64_t)H, grid);
+  const size_t bwd_lds = (size_t)h_slice * 4 * H * sizeof(float);
+  TORCH_CHECK(bwd_lds <= 64 * 1024, "lstm hidden size too large for LDS slices");
   {
     int cap = coop_grid_size((const void*)lstm_bwd_kernel, threads, bwd_lds);
     TORCH_CHECK(grid <= cap, "lstm bwd grid exceeds cooperative capacity");
