@@ -494,13 +494,13 @@ def test(flags, num_episodes: int = 10):
     model.load_state_dict(checkpoint["model_state_dict"])
 
     observation = env.initial()
+    agent_state = model.initial_state(batch_size=1)
     returns = []
 
     while len(returns) < num_episodes:
         if flags.mode == "test_render":
             env.env.render()
-        agent_outputs = model(observation)
-        policy_outputs, _ = agent_outputs
+        policy_outputs, agent_state = model(observation, agent_state)
         observation = env.step(policy_outputs["action"])
         if observation["done"].item():
             returns.append(observation["episode_return"].item())
