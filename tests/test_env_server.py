@@ -186,3 +186,43 @@ def test_server_socket_file_cleanup():
     assert os.path.exists(path)
     server.stop()
     assert not os.path.exists(path)
+
+
+class MultiDtypeEnv:
+    """Fixture covering the wire codec's dtype table: one observation array
+    of the parametrized dtype, offset per step so values are checkable."""
+
+    def __init__(self, dtype):
+        self.dtype = dtype
+        self._t = 0
+
+    def reset(self):
+        self._t = 0
+        return np.arange(6, dtype=self.dtype).reshape(2, 3)
+
+    def step(self, action):
+        self._t += 1
+        obs = (np.arange(6, dtype=self.dtype).reshape(2, 3)
+               + np.asarray(self._t, dtype=self.dtype))
+        return obs, 0.5, False, {}
+
+
+@pytest.mark.parametrize("dtype", [np.uint8, np.int16, np.int32, np.int64,
+                                   np.float32, np.float64])
+def test_wire_roundtrip_dtypes(dtype):
+    addr = _address()
+    server = runtime.Server(lambda: MultiDtypeEnv(dtype), addr)
+    server.start()
+    try:
+        def policy(env_outputs, agent_state):
+            b = env_outputs[0].shape[1]
+            return ((torch.zeros(1, b, dtype=torch.int64),
+                     torch.zeros(1, b, 2), torch.zeros(1, b)), agent_state)
+
+        rollouts = _run_pool(2, [addr], policy, n_rollouts=1)
+    finally:
+        server.stop()
+    frames = rollouts[0][0][0][0]  # [T+1, 1, 2, 3]
+    expected0 = np.arange(6, dtype=dtype).reshape(2, 3)
+    np.testing.assert_array_equal(frames[0, 0].numpy(), expected0)
+    np.testing.assert_array_equal(frames[1, 0].numpy(), expected0 + 1)

@@ -118,3 +118,36 @@ def test_polybeast_test_mode(tmp_path):
     polybeast_learner.train(flags)
     avg = polybeast_learner.test(flags, num_episodes=1)
     assert isinstance(avg, float)
+
+
+def test_train_two_learner_threads(tmp_path):
+    """Two learner threads share the model under the learn lock (the
+    reference's default num_learner_threads=2 topology)."""
+    flags = _flags(tmp_path, num_learner_threads=2, total_steps=96)
+    polybeast_learner.train(flags)
+    ckpt = torch.load(
+        os.path.join(str(tmp_path), "pbtest", "model.tar"),
+        map_location="cpu", weights_only=False,
+    )
+    assert ckpt["stats"]["step"] >= 96
+
+
+def test_train_lstm_over_socket_servers(tmp_path):
+    """Recurrent state round-trips through the wire protocol + batcher while
+    training against env servers."""
+    from torchbeast_amd import polybeast_env, runtime
+
+    basename = f"unix:/tmp/tbamd-pbl-{uuid.uuid4().hex[:8]}"
+    factory = polybeast_env.create_env_factory("synthetic:4x84x84:6")
+    servers = [runtime.Server(factory, f"{basename}.{i}") for i in range(2)]
+    for s in servers:
+        s.start()
+    try:
+        flags = _flags(tmp_path, num_actors=2, pipes_basename=basename,
+                       total_steps=32, use_lstm=True)
+        flags.env = "PongNoFrameskip-v4"  # socket addressing path
+        flags.num_actions = 6
+        polybeast_learner.train(flags)
+    finally:
+        for s in servers:
+            s.stop()
