@@ -1,19 +1,19 @@
-"""Online per-phase timing statistics (ref: torchbeast/core/prof.py).
+"""Per-phase wall-time statistics for hot loops (capability parity with the
+reference's Timings helper, ref: torchbeast/core/prof.py).
 
-`Timings.time(name)` records the elapsed wall time since the previous call
-and folds it into a running mean/variance (Welford). `summary()` renders
-per-phase means, standard deviations and percentage of total.
+Usage: call `time(name)` after each phase of a loop body; the elapsed time
+since the previous call is folded into an online mean/variance (Welford's
+algorithm, one (count, mean, M2) triple per phase name). `summary()` renders
+the phases sorted by cost with their share of the total.
 """
 
-import collections
 import timeit
 
 
 class Timings:
     def __init__(self):
-        self._means = collections.defaultdict(int)
-        self._vars = collections.defaultdict(int)
-        self._counts = collections.defaultdict(int)
+        # name -> [count, mean, M2]  (Welford accumulator)
+        self._acc = {}
         self.reset()
 
     def reset(self):
@@ -21,40 +21,37 @@ class Timings:
 
     def time(self, name):
         now = timeit.default_timer()
-        x = now - self.last_time
+        sample = now - self.last_time
         self.last_time = now
 
-        n = self._counts[name]
-        mean = self._means[name] + (x - self._means[name]) / (n + 1)
-        var = (
-            n * self._vars[name] + n * (self._means[name] - mean) ** 2 + (x - mean) ** 2
-        ) / (n + 1)
-
-        self._means[name] = mean
-        self._vars[name] = var
-        self._counts[name] += 1
+        entry = self._acc.setdefault(name, [0, 0.0, 0.0])
+        entry[0] += 1
+        delta = sample - entry[1]
+        entry[1] += delta / entry[0]
+        entry[2] += delta * (sample - entry[1])
 
     def means(self):
-        return self._means
+        return {name: e[1] for name, e in self._acc.items()}
 
     def vars(self):
-        return self._vars
+        return {
+            name: (e[2] / e[0] if e[0] else 0.0) for name, e in self._acc.items()
+        }
 
     def stds(self):
-        return {k: v**0.5 for k, v in self._vars.items()}
+        return {name: v**0.5 for name, v in self.vars().items()}
 
     def summary(self, prefix=""):
         means = self.means()
         stds = self.stds()
         total = sum(means.values())
 
-        result = prefix
-        for k in sorted(means, key=means.get, reverse=True):
-            result += "\n    %s: %.6fms +- %.6fms (%.2f%%) " % (
-                k,
-                1000 * means[k],
-                1000 * stds[k],
-                100 * means[k] / total if total else 0.0,
+        lines = [prefix]
+        for name in sorted(means, key=means.get, reverse=True):
+            share = 100 * means[name] / total if total else 0.0
+            lines.append(
+                f"    {name}: {1000 * means[name]:.6f}ms "
+                f"+- {1000 * stds[name]:.6f}ms ({share:.2f}%) "
             )
-        result += "\nTotal: %.6fms" % (1000 * total)
-        return result
+        lines.append(f"Total: {1000 * total:.6f}ms")
+        return "\n".join(lines)

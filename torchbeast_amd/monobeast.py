@@ -1,15 +1,15 @@
-"""MonoBeast: single-machine IMPALA (ref: torchbeast/monobeast.py).
+"""MonoBeast: single-machine IMPALA (capability parity with
+torchbeast/monobeast.py — fresh MI355X-native implementation).
 
-Actors are forked processes, each stepping its own environment and a shared
-CPU copy of the model; rollouts travel through shared-memory tensor buffers
-indexed by free/full queues; learner threads stack buffers into [T+1, B]
-batches, move them to the GPU, and run V-trace + losses + RMSProp.
+Topology: forked actor processes each own an environment plus the shared
+CPU behavior model and fill shared-memory rollout slots, handing slot ids
+through free/full queues; learner threads stack slots into [T+1, B] batches,
+push them to the GPU and run V-trace + IMPALA losses + RMSProp, then copy
+fresh weights back into the shared behavior model.
 
-MI355X specifics relative to the reference:
-- On ROCm devices the V-trace scan, the fused loss and the LSTM unroll
-  dispatch to CDNA4 HIP kernels (torchbeast_amd/ops/).
-- Batches are staged through pinned host memory so the H2D copy is a DMA on
-  a side stream (`--pin_buffers`).
+MI355X specifics: on ROCm devices the V-trace scan, the fused loss, the
+conv trunk and the LSTM unroll dispatch to the CDNA4 HIP kernels in
+torchbeast_amd/ops/.
 
 Run: python -m torchbeast_amd.monobeast --env synthetic --num_actors 4
 """
@@ -33,7 +33,7 @@ from torchbeast_amd.core import environment, file_writer, prof, vtrace
 from torchbeast_amd.models.atari_net import AtariNet
 from torchbeast_amd.ops import functional as tbops
 
-# Re-exported for API parity with the reference module.
+# The reference exports its model under both names; keep that surface.
 Net = AtariNet
 
 logging.basicConfig(
@@ -42,6 +42,9 @@ logging.basicConfig(
 )
 
 Buffers = typing.Dict[str, typing.List[torch.Tensor]]
+
+CHECKPOINT_INTERVAL_S = 10 * 60
+MONITOR_INTERVAL_S = 5
 
 
 def make_parser():
@@ -83,6 +86,42 @@ def create_env(flags):
     )
 
 
+def _env_interface(env):
+    """(obs_shape, num_actions) for either gym-style or synthetic envs."""
+    if hasattr(env, "num_actions"):
+        return env.reset().shape, env.num_actions
+    return env.observation_space.shape, env.action_space.n
+
+
+def _checkpoint_path(flags):
+    return os.path.expandvars(
+        os.path.expanduser(f"{flags.savedir}/{flags.xpid}/model.tar")
+    )
+
+
+def create_buffers(flags, obs_shape, num_actions) -> Buffers:
+    """Shared-memory rollout slots: one [T+1, ...] tensor per field per slot."""
+    T1 = flags.unroll_length + 1
+    field_specs = [
+        ("frame", (T1, *obs_shape), torch.uint8),
+        ("reward", (T1,), torch.float32),
+        ("done", (T1,), torch.bool),
+        ("episode_return", (T1,), torch.float32),
+        ("episode_step", (T1,), torch.int32),
+        ("policy_logits", (T1, num_actions), torch.float32),
+        ("baseline", (T1,), torch.float32),
+        ("last_action", (T1,), torch.int64),
+        ("action", (T1,), torch.int64),
+    ]
+    buffers: Buffers = {}
+    for name, shape, dtype in field_specs:
+        buffers[name] = [
+            torch.empty(shape, dtype=dtype).share_memory_()
+            for _ in range(flags.num_buffers)
+        ]
+    return buffers
+
+
 def act(
     flags,
     actor_index: int,
@@ -92,63 +131,56 @@ def act(
     buffers: Buffers,
     initial_agent_state_buffers,
 ):
-    """Actor process: step the env with the shared behavior model, writing
-    each unroll into the shared-memory buffer slot it takes from free_queue."""
+    """Actor-process main: run the env with the shared behavior model and
+    fill whichever rollout slot free_queue hands out."""
     try:
-        # Actors are forked: cap intra-op threads BEFORE the first tensor op
-        # (a multi-threaded OpenMP pool inherited across fork deadlocks), and
-        # one thread per actor is the right sizing anyway.
+        # We are a fork: pin intra-op threads to 1 BEFORE any tensor op (an
+        # inherited multi-threaded OpenMP pool deadlocks across fork), and
+        # a single thread per actor is the right sizing anyway.
         torch.set_num_threads(1)
-        logging.info("Actor %i started.", actor_index)
+        logging.info("Actor %d up.", actor_index)
         timings = prof.Timings()
 
-        gym_env = create_env(flags)
-        if hasattr(gym_env, "seed"):
-            seed = actor_index ^ int.from_bytes(os.urandom(4), byteorder="little")
-            gym_env.seed(seed)
-        env = environment.Environment(gym_env)
-        env_output = env.initial()
+        raw_env = create_env(flags)
+        if hasattr(raw_env, "seed"):
+            raw_env.seed(actor_index ^ int.from_bytes(os.urandom(4), "little"))
+        env = environment.Environment(raw_env)
+
+        step_out = env.initial()
         agent_state = model.initial_state(batch_size=1)
-        agent_output, unused_state = model(env_output, agent_state)
+        agent_out, _ = model(step_out, agent_state)
 
-        while True:
-            index = free_queue.get()
-            if index is None:
-                break
+        def record(slot, t):
+            for name, value in step_out.items():
+                buffers[name][slot][t, ...] = value
+            for name, value in agent_out.items():
+                buffers[name][slot][t, ...] = value
 
-            # Step 0 of the new rollout is the final step of the previous one.
-            for key in env_output:
-                buffers[key][index][0, ...] = env_output[key]
-            for key in agent_output:
-                buffers[key][index][0, ...] = agent_output[key]
-            for i, t in enumerate(agent_state):
-                initial_agent_state_buffers[index][i][...] = t
+        while (slot := free_queue.get()) is not None:
+            # Rollouts overlap by one step: row 0 repeats the previous
+            # rollout's final step; the recurrent state snapshot goes with it.
+            record(slot, 0)
+            for i, s in enumerate(agent_state):
+                initial_agent_state_buffers[slot][i][...] = s
 
             for t in range(flags.unroll_length):
                 timings.reset()
-
                 with torch.no_grad():
-                    agent_output, agent_state = model(env_output, agent_state)
+                    agent_out, agent_state = model(step_out, agent_state)
                 timings.time("model")
-
-                env_output = env.step(agent_output["action"])
+                step_out = env.step(agent_out["action"])
                 timings.time("step")
-
-                for key in env_output:
-                    buffers[key][index][t + 1, ...] = env_output[key]
-                for key in agent_output:
-                    buffers[key][index][t + 1, ...] = agent_output[key]
+                record(slot, t + 1)
                 timings.time("write")
 
-            full_queue.put(index)
+            full_queue.put(slot)
 
         if actor_index == 0:
             logging.info("Actor 0 timings: %s", timings.summary())
-
     except KeyboardInterrupt:
-        pass  # Silently exit on ctrl-c; the main process handles shutdown.
+        pass  # Parent coordinates shutdown.
     except Exception:
-        logging.error("Exception in actor %i:\n%s", actor_index, traceback.format_exc())
+        logging.error("Actor %d died:\n%s", actor_index, traceback.format_exc())
         raise
 
 
@@ -161,24 +193,32 @@ def get_batch(
     timings,
     lock=threading.Lock(),
 ):
+    """Claim batch_size full slots, stack them along dim 1, release the
+    slots, and move everything to the learner device."""
     with lock:
         timings.time("lock")
-        indices = [full_queue.get() for _ in range(flags.batch_size)]
+        slots = [full_queue.get() for _ in range(flags.batch_size)]
         timings.time("dequeue")
+
     batch = {
-        key: torch.stack([buffers[key][m] for m in indices], dim=1) for key in buffers
+        name: torch.stack([column[s] for s in slots], dim=1)
+        for name, column in buffers.items()
     }
-    initial_agent_state = (
-        torch.cat(ts, dim=1)
-        for ts in zip(*[initial_agent_state_buffers[m] for m in indices])
-    )
+    state_columns = zip(*(initial_agent_state_buffers[s] for s in slots))
+    initial_agent_state = [torch.cat(col, dim=1) for col in state_columns]
     timings.time("batch")
-    for m in indices:
-        free_queue.put(m)
+
+    for s in slots:
+        free_queue.put(s)
     timings.time("enqueue")
-    batch = {k: t.to(device=flags.device, non_blocking=True) for k, t in batch.items()}
+
+    batch = {
+        name: t.to(device=flags.device, non_blocking=True)
+        for name, t in batch.items()
+    }
     initial_agent_state = tuple(
-        t.to(device=flags.device, non_blocking=True) for t in initial_agent_state
+        t.to(device=flags.device, non_blocking=True)
+        for t in initial_agent_state
     )
     timings.time("device")
     return batch, initial_agent_state
@@ -194,32 +234,30 @@ def learn(
     scheduler,
     lock=threading.Lock(),
 ):
-    """One learner step: forward, V-trace targets, losses, backward, RMSProp,
-    and a weight push back to the shared behavior model."""
+    """One optimizer step on a [T+1, B] batch, then a weight push to the
+    shared behavior model."""
     with lock:
-        learner_outputs, unused_state = model(batch, initial_agent_state)
+        learner_outputs, _ = model(batch, initial_agent_state)
 
-        # Bootstrap from the value estimate at the final step.
+        # The value estimate at the last row bootstraps the returns.
         bootstrap_value = learner_outputs["baseline"][-1]
 
-        # Align: env_output[t+1] is the consequence of agent_output[t].
-        batch = {key: tensor[1:] for key, tensor in batch.items()}
-        learner_outputs = {key: tensor[:-1] for key, tensor in learner_outputs.items()}
+        # Re-align so that row t pairs action[t] with its consequence.
+        batch = {name: t[1:] for name, t in batch.items()}
+        learner_outputs = {name: t[:-1] for name, t in learner_outputs.items()}
 
-        rewards = batch["reward"]
         if flags.reward_clipping == "abs_one":
-            clipped_rewards = torch.clamp(rewards, -1, 1)
+            rewards = torch.clamp(batch["reward"], -1, 1)
         else:
-            clipped_rewards = rewards
-
+            rewards = batch["reward"]
         discounts = (~batch["done"]).float() * flags.discounting
 
-        vtrace_returns = vtrace.from_logits(
+        targets = vtrace.from_logits(
             behavior_policy_logits=batch["policy_logits"],
             target_policy_logits=learner_outputs["policy_logits"],
             actions=batch["action"],
             discounts=discounts,
-            rewards=clipped_rewards,
+            rewards=rewards,
             values=learner_outputs["baseline"],
             bootstrap_value=bootstrap_value,
         )
@@ -228,8 +266,8 @@ def learn(
             learner_outputs["policy_logits"],
             learner_outputs["baseline"],
             batch["action"],
-            vtrace_returns.pg_advantages,
-            vtrace_returns.vs,
+            targets.pg_advantages,
+            targets.vs,
         )
         total_loss = (
             pg_loss
@@ -237,10 +275,10 @@ def learn(
             + flags.entropy_cost * entropy_loss
         )
 
-        episode_returns = batch["episode_return"][batch["done"]]
+        finished = batch["episode_return"][batch["done"]]
         stats = {
-            "episode_returns": tuple(episode_returns.tolist()),
-            "mean_episode_return": torch.mean(episode_returns).item(),
+            "episode_returns": tuple(finished.tolist()),
+            "mean_episode_return": torch.mean(finished).item(),
             "total_loss": total_loss.item(),
             "pg_loss": pg_loss.item(),
             "baseline_loss": baseline_loss.item(),
@@ -257,34 +295,11 @@ def learn(
         return stats
 
 
-def create_buffers(flags, obs_shape, num_actions) -> Buffers:
-    T = flags.unroll_length
-    specs = dict(
-        frame=dict(size=(T + 1, *obs_shape), dtype=torch.uint8),
-        reward=dict(size=(T + 1,), dtype=torch.float32),
-        done=dict(size=(T + 1,), dtype=torch.bool),
-        episode_return=dict(size=(T + 1,), dtype=torch.float32),
-        episode_step=dict(size=(T + 1,), dtype=torch.int32),
-        policy_logits=dict(size=(T + 1, num_actions), dtype=torch.float32),
-        baseline=dict(size=(T + 1,), dtype=torch.float32),
-        last_action=dict(size=(T + 1,), dtype=torch.int64),
-        action=dict(size=(T + 1,), dtype=torch.int64),
-    )
-    buffers: Buffers = {key: [] for key in specs}
-    for _ in range(flags.num_buffers):
-        for key in buffers:
-            buffers[key].append(torch.empty(**specs[key]).share_memory_())
-    return buffers
-
-
 def train(flags):  # noqa: C901
     if flags.xpid is None:
-        flags.xpid = "torchbeast-%s" % time.strftime("%Y%m%d-%H%M%S")
+        flags.xpid = time.strftime("torchbeast-%Y%m%d-%H%M%S")
     plogger = file_writer.FileWriter(
         xpid=flags.xpid, xp_args=flags.__dict__, rootdir=flags.savedir
-    )
-    checkpointpath = os.path.expandvars(
-        os.path.expanduser("%s/%s/%s" % (flags.savedir, flags.xpid, "model.tar"))
     )
 
     if flags.num_buffers is None:
@@ -294,62 +309,44 @@ def train(flags):  # noqa: C901
     if flags.num_buffers < flags.batch_size:
         raise ValueError("num_buffers should be larger than batch_size")
 
-    T = flags.unroll_length
-    B = flags.batch_size
+    T, B = flags.unroll_length, flags.batch_size
+    steps_per_update = T * B
 
-    flags.device = None
-    if not flags.disable_cuda and torch.cuda.is_available():
-        logging.info("Using CUDA (ROCm).")
-        flags.device = torch.device("cuda")
-    else:
-        logging.info("Not using CUDA.")
-        flags.device = torch.device("cpu")
+    use_gpu = not flags.disable_cuda and torch.cuda.is_available()
+    flags.device = torch.device("cuda" if use_gpu else "cpu")
+    logging.info("Learner device: %s", flags.device)
 
-    env = create_env(flags)
-    obs_shape = env.reset().shape if hasattr(env, "reset") else env.observation_space.shape
-    num_actions = (
-        env.num_actions
-        if hasattr(env, "num_actions")
-        else env.action_space.n
-    )
+    probe_env = create_env(flags)
+    obs_shape, num_actions = _env_interface(probe_env)
 
     model = Net(obs_shape, num_actions, flags.use_lstm)
+    model.share_memory()
     buffers = create_buffers(flags, obs_shape, model.num_actions)
 
-    model.share_memory()
-
-    # Shared slots for the recurrent state at each rollout's first step.
+    # One shared recurrent-state snapshot per rollout slot.
     initial_agent_state_buffers = []
     for _ in range(flags.num_buffers):
-        state = model.initial_state(batch_size=1)
-        for t in state:
-            t.share_memory_()
-        initial_agent_state_buffers.append(state)
+        snapshot = model.initial_state(batch_size=1)
+        for s in snapshot:
+            s.share_memory_()
+        initial_agent_state_buffers.append(snapshot)
 
-    actor_processes = []
     ctx = mp.get_context("fork")
     free_queue = ctx.SimpleQueue()
     full_queue = ctx.SimpleQueue()
 
+    actor_processes = []
     for i in range(flags.num_actors):
-        actor = ctx.Process(
+        p = ctx.Process(
             target=act,
-            args=(
-                flags,
-                i,
-                free_queue,
-                full_queue,
-                model,
-                buffers,
-                initial_agent_state_buffers,
-            ),
+            args=(flags, i, free_queue, full_queue, model, buffers,
+                  initial_agent_state_buffers),
         )
-        actor.start()
-        actor_processes.append(actor)
+        p.start()
+        actor_processes.append(p)
 
-    learner_model = Net(obs_shape, num_actions, flags.use_lstm).to(
-        device=flags.device
-    )
+    learner_model = Net(obs_shape, num_actions, flags.use_lstm)
+    learner_model = learner_model.to(device=flags.device)
     learner_model.load_state_dict(model.state_dict())
 
     optimizer = torch.optim.RMSprop(
@@ -359,66 +356,55 @@ def train(flags):  # noqa: C901
         eps=flags.epsilon,
         alpha=flags.alpha,
     )
+    # Linear decay to zero over total_steps (in env steps).
+    scheduler = torch.optim.lr_scheduler.LambdaLR(
+        optimizer,
+        lambda n_updates: 1
+        - min(n_updates * steps_per_update, flags.total_steps)
+        / flags.total_steps,
+    )
 
-    def lr_lambda(epoch):
-        return 1 - min(epoch * T * B, flags.total_steps) / flags.total_steps
-
-    scheduler = torch.optim.lr_scheduler.LambdaLR(optimizer, lr_lambda)
-
-    logger = logging.getLogger("logfile")
-    stat_keys = [
-        "total_loss",
-        "mean_episode_return",
-        "pg_loss",
-        "baseline_loss",
-        "entropy_loss",
-    ]
-    logger.info("# Step\t%s", "\t".join(stat_keys))
+    stat_keys = ["total_loss", "mean_episode_return", "pg_loss",
+                 "baseline_loss", "entropy_loss"]
+    logging.getLogger("logfile").info("# Step\t%s", "\t".join(stat_keys))
 
     step, stats = 0, {}
 
-    def batch_and_learn(i, lock=threading.Lock()):
-        """Learner thread: repeatedly assemble a batch and run learn()."""
+    def learner_thread_main(thread_id, log_lock=threading.Lock()):
         nonlocal step, stats
         timings = prof.Timings()
         while step < flags.total_steps:
             timings.reset()
             batch, agent_state = get_batch(
-                flags,
-                free_queue,
-                full_queue,
-                buffers,
-                initial_agent_state_buffers,
-                timings,
+                flags, free_queue, full_queue, buffers,
+                initial_agent_state_buffers, timings,
             )
-            stats = learn(
-                flags, model, learner_model, batch, agent_state, optimizer, scheduler
-            )
+            stats = learn(flags, model, learner_model, batch, agent_state,
+                          optimizer, scheduler)
             timings.time("learn")
-            with lock:
-                to_log = dict(step=step)
-                to_log.update({k: stats[k] for k in stat_keys})
-                plogger.log(to_log)
-                step += T * B
+            with log_lock:
+                row = {"step": step, **{k: stats[k] for k in stat_keys}}
+                plogger.log(row)
+                step += steps_per_update
+        if thread_id == 0:
+            logging.info("Learner timings: %s", timings.summary())
 
-        if i == 0:
-            logging.info("Batch and learn: %s", timings.summary())
+    for slot in range(flags.num_buffers):
+        free_queue.put(slot)
 
-    for m in range(flags.num_buffers):
-        free_queue.put(m)
-
-    threads = []
-    for i in range(flags.num_learner_threads):
-        thread = threading.Thread(
-            target=batch_and_learn, name="batch-and-learn-%d" % i, args=(i,)
-        )
-        thread.start()
-        threads.append(thread)
+    threads = [
+        threading.Thread(target=learner_thread_main, args=(i,),
+                         name=f"learner-{i}")
+        for i in range(flags.num_learner_threads)
+    ]
+    for t in threads:
+        t.start()
 
     def checkpoint():
         if flags.disable_checkpoint:
             return
-        logging.info("Saving checkpoint to %s", checkpointpath)
+        path = _checkpoint_path(flags)
+        logging.info("Checkpointing to %s", path)
         torch.save(
             {
                 "model_state_dict": learner_model.state_dict(),
@@ -426,48 +412,41 @@ def train(flags):  # noqa: C901
                 "scheduler_state_dict": scheduler.state_dict(),
                 "flags": vars(flags),
             },
-            checkpointpath,
+            path,
         )
 
-    timer = timeit.default_timer
+    clock = timeit.default_timer
     try:
-        last_checkpoint_time = timer()
+        last_ckpt = clock()
         while step < flags.total_steps:
-            start_step = step
-            start_time = timer()
-            time.sleep(5)
+            window_start_step, window_start_time = step, clock()
+            time.sleep(MONITOR_INTERVAL_S)
 
-            if timer() - last_checkpoint_time > 10 * 60:
+            if clock() - last_ckpt > CHECKPOINT_INTERVAL_S:
                 checkpoint()
-                last_checkpoint_time = timer()
+                last_ckpt = clock()
 
-            sps = (step - start_step) / (timer() - start_time)
-            if stats.get("episode_returns", None):
-                mean_return = (
-                    "Return per episode: %.1f. " % stats["mean_episode_return"]
-                )
+            sps = (step - window_start_step) / (clock() - window_start_time)
+            if stats.get("episode_returns"):
+                ret_info = f"Return per episode: {stats['mean_episode_return']:.1f}. "
             else:
-                mean_return = ""
-            total_loss = stats.get("total_loss", float("inf"))
+                ret_info = ""
             logging.info(
                 "Steps %i @ %.1f SPS. Loss %f. %sStats:\n%s",
-                step,
-                sps,
-                total_loss,
-                mean_return,
+                step, sps, stats.get("total_loss", float("inf")), ret_info,
                 pprint.pformat(stats),
             )
     except KeyboardInterrupt:
-        return  # Try joining actors then quit.
+        return  # Fall through to finally for shutdown.
     else:
-        for thread in threads:
-            thread.join()
+        for t in threads:
+            t.join()
         logging.info("Learning finished after %d steps.", step)
     finally:
         for _ in range(flags.num_actors):
             free_queue.put(None)
-        for actor in actor_processes:
-            actor.join(timeout=1)
+        for p in actor_processes:
+            p.join(timeout=1)
         checkpoint()
         plogger.close()
 
@@ -476,27 +455,21 @@ def test(flags, num_episodes: int = 10):
     if flags.xpid is None:
         checkpointpath = "./latest/model.tar"
     else:
-        checkpointpath = os.path.expandvars(
-            os.path.expanduser("%s/%s/%s" % (flags.savedir, flags.xpid, "model.tar"))
-        )
+        checkpointpath = _checkpoint_path(flags)
 
-    gym_env = create_env(flags)
-    env = environment.Environment(gym_env)
-    obs_shape = gym_env.reset().shape if hasattr(gym_env, "reset") else gym_env.observation_space.shape
-    num_actions = (
-        gym_env.num_actions
-        if hasattr(gym_env, "num_actions")
-        else gym_env.action_space.n
-    )
+    raw_env = create_env(flags)
+    obs_shape, num_actions = _env_interface(raw_env)
+    env = environment.Environment(raw_env)
+
     model = Net(obs_shape, num_actions, flags.use_lstm)
-    model.eval()
-    checkpoint = torch.load(checkpointpath, map_location="cpu", weights_only=False)
-    model.load_state_dict(checkpoint["model_state_dict"])
+    model.eval()  # Greedy action selection.
+    snapshot = torch.load(checkpointpath, map_location="cpu",
+                          weights_only=False)
+    model.load_state_dict(snapshot["model_state_dict"])
 
     observation = env.initial()
     agent_state = model.initial_state(batch_size=1)
     returns = []
-
     while len(returns) < num_episodes:
         if flags.mode == "test_render":
             env.env.render()
@@ -510,9 +483,8 @@ def test(flags, num_episodes: int = 10):
                 observation["episode_return"].item(),
             )
     env.close()
-    logging.info(
-        "Average returns over %i episodes: %.1f", num_episodes, sum(returns) / len(returns)
-    )
+    logging.info("Average returns over %i episodes: %.1f",
+                 num_episodes, sum(returns) / len(returns))
 
 
 def main(flags):
