@@ -40,7 +40,11 @@ ext_modules = [
     )
 ]
 
-hip_sources = sorted(glob.glob("torchbeast_amd/ops/hip/*.hip"))
+# torch's hipify writes *_hip.hip shadow copies next to the sources; they
+# define the same symbols, so they must never enter the source list.
+hip_sources = sorted(
+    p for p in glob.glob("torchbeast_amd/ops/hip/*.hip") if not p.endswith("_hip.hip")
+)
 if hip_sources:
     ext_modules.append(
         CUDAExtension(

@@ -1,13 +1,18 @@
 """Experiment metrics/metadata writer (ref: torchbeast/core/file_writer.py).
 
-Creates `{rootdir}/{xpid}/` containing:
-- `out.log`     — log messages,
-- `logs.csv`    — one row per `log()` tick; the column schema grows
-                  dynamically as new metric keys appear,
-- `fields.csv`  — history of the column schema,
-- `meta.json`   — experiment args + environment metadata,
-and maintains a `latest` symlink next to the xpid directory. On re-creation
-with an existing xpid the writer appends, continuing the tick counter.
+One directory per experiment id under `rootdir`:
+
+    {rootdir}/{xpid}/
+        out.log      stream of human-readable messages
+        logs.csv     one row per log() call, tick-indexed; columns grow
+                     as new metric names show up
+        fields.csv   one row per schema revision (audit trail of columns)
+        meta.json    experiment args + host/git/scheduler metadata
+    {rootdir}/latest -> the most recently created xpid directory
+
+Re-opening an existing xpid appends instead of truncating and resumes the
+tick counter from the last row, so a preempted-and-restarted run produces
+one continuous CSV.
 """
 
 import copy
@@ -16,128 +21,129 @@ import datetime
 import json
 import logging
 import os
+import subprocess
 import time
-from typing import Dict
+from typing import Dict, List, Optional
 
 
-def _gather_metadata() -> Dict:
-    date_start = datetime.datetime.now().isoformat()
-    # Git metadata, when running from a checkout.
-    git = {}
+def _git_commit() -> Optional[str]:
     try:
-        import subprocess
-
-        git["commit"] = (
-            subprocess.check_output(
-                ["git", "rev-parse", "HEAD"], stderr=subprocess.DEVNULL
-            )
-            .decode()
-            .strip()
+        out = subprocess.run(
+            ["git", "rev-parse", "HEAD"],
+            capture_output=True,
+            timeout=5,
+            check=True,
         )
+        return out.stdout.decode().strip()
     except Exception:
-        pass
-    slurm = {
-        k.split("_", 1)[1].lower(): v
-        for k, v in os.environ.items()
-        if k.startswith("SLURM_")
-    }
-    return dict(
-        date_start=date_start,
-        date_end=None,
-        successful=False,
-        git=git,
-        slurm=slurm or None,
-        env=dict(os.environ),
-    )
+        return None
+
+
+def _scheduler_metadata() -> Optional[Dict[str, str]]:
+    """SLURM job context, if any (lowercased keys without the prefix)."""
+    found = {}
+    for key, value in os.environ.items():
+        if not key.startswith("SLURM_"):
+            continue
+        found[key[len("SLURM_"):].lower()] = value
+    return found if found else None
 
 
 class FileWriter:
     def __init__(self, xpid: str = None, xp_args: dict = None, rootdir: str = "~/palaas"):
-        if not xpid:
-            xpid = f"{os.getpid()}_{int(time.time())}"
-        self.xpid = xpid
+        self.xpid = xpid or f"{os.getpid()}_{int(time.time())}"
         self._tick = 0
 
-        self.metadata = _gather_metadata()
-        self.metadata["args"] = copy.deepcopy(xp_args) if xp_args else {}
-        self.metadata["xpid"] = self.xpid
-
-        formatter = logging.Formatter("%(message)s")
-        self._logger = logging.getLogger(f"palaas/{xpid}")
-        self._logger.propagate = False
-        self._logger.setLevel(logging.INFO)
-        if not self._logger.hasHandlers():
-            self._logger.addHandler(logging.StreamHandler())
+        self.metadata = {
+            "xpid": self.xpid,
+            "args": copy.deepcopy(xp_args) if xp_args else {},
+            "date_start": datetime.datetime.now().isoformat(),
+            "date_end": None,
+            "successful": False,
+            "git": {"commit": _git_commit()},
+            "slurm": _scheduler_metadata(),
+            "env": dict(os.environ),
+        }
 
         rootdir = os.path.expandvars(os.path.expanduser(rootdir))
         self.basepath = os.path.join(rootdir, self.xpid)
         os.makedirs(self.basepath, exist_ok=True)
+        self._repoint_latest_symlink(rootdir)
 
-        # Point {rootdir}/latest at the newest experiment directory.
-        symlink = os.path.join(rootdir, "latest")
-        try:
-            if os.path.islink(symlink):
-                os.remove(symlink)
-            if not os.path.exists(symlink):
-                os.symlink(self.basepath, symlink)
-        except OSError:
-            pass
+        self.paths = {
+            name: os.path.join(self.basepath, filename)
+            for name, filename in (
+                ("msg", "out.log"),
+                ("logs", "logs.csv"),
+                ("fields", "fields.csv"),
+                ("meta", "meta.json"),
+            )
+        }
 
-        self.paths = dict(
-            msg=os.path.join(self.basepath, "out.log"),
-            logs=os.path.join(self.basepath, "logs.csv"),
-            fields=os.path.join(self.basepath, "fields.csv"),
-            meta=os.path.join(self.basepath, "meta.json"),
-        )
-
+        self._logger = logging.getLogger(f"palaas/{self.xpid}")
+        self._logger.propagate = False
+        self._logger.setLevel(logging.INFO)
+        if not self._logger.hasHandlers():
+            self._logger.addHandler(logging.StreamHandler())
         self._logger.addHandler(logging.FileHandler(self.paths["msg"]))
+        plain = logging.Formatter("%(message)s")
         for handler in self._logger.handlers:
-            handler.setFormatter(formatter)
+            handler.setFormatter(plain)
 
         self._save_metadata()
+        self.fieldnames = self._recover_schema()
 
-        self.fieldnames = ["_tick", "_time"]
-        if os.path.exists(self.paths["logs"]):
-            # Resume: recover the schema and continue the tick counter.
-            with open(self.paths["logs"], "r") as f:
-                reader = csv.reader(f)
-                lines = list(reader)
-            if len(lines) > 1:
-                self.fieldnames = lines[0]
-                try:
-                    self._tick = int(lines[-1][0]) + 1
-                except (ValueError, IndexError):
-                    pass
+    def _repoint_latest_symlink(self, rootdir: str) -> None:
+        link = os.path.join(rootdir, "latest")
+        try:
+            if os.path.islink(link):
+                os.remove(link)
+            if not os.path.exists(link):
+                os.symlink(self.basepath, link)
+        except OSError:
+            pass  # e.g. filesystems without symlink support
+
+    def _recover_schema(self) -> List[str]:
+        """On resume, re-read logs.csv for the column list and last tick."""
+        default = ["_tick", "_time"]
+        if not os.path.exists(self.paths["logs"]):
+            return default
+        with open(self.paths["logs"], "r") as f:
+            lines = list(csv.reader(f))
+        if len(lines) < 2:
+            return default
+        try:
+            self._tick = int(lines[-1][0]) + 1
+        except (ValueError, IndexError):
+            pass
+        return lines[0]
 
     def log(self, to_log: Dict, tick: int = None, verbose: bool = False) -> None:
         if tick is not None:
-            raise NotImplementedError
+            raise NotImplementedError("explicit ticks are not supported")
         to_log["_tick"] = self._tick
-        self._tick += 1
         to_log["_time"] = time.time()
+        self._tick += 1
 
-        old_len = len(self.fieldnames)
-        for k in to_log:
-            if k not in self.fieldnames:
-                self.fieldnames.append(k)
-        if old_len != len(self.fieldnames):
+        new_columns = [k for k in to_log if k not in self.fieldnames]
+        if new_columns:
+            self.fieldnames.extend(new_columns)
             with open(self.paths["fields"], "a") as f:
                 csv.writer(f).writerow(self.fieldnames)
             self._logger.info("Updated log fields: %s", self.fieldnames)
 
         if to_log["_tick"] == 0:
+            # Header row, written once. Commented so pandas/np loaders that
+            # skip '#' lines and the resume parser both cope.
             with open(self.paths["logs"], "a") as f:
                 f.write("# %s\n" % ",".join(self.fieldnames))
 
         if verbose:
-            self._logger.info(
-                "LOG | %s",
-                ", ".join(f"{k}: {to_log[k]}" for k in sorted(to_log)),
-            )
+            rendered = ", ".join(f"{k}: {to_log[k]}" for k in sorted(to_log))
+            self._logger.info("LOG | %s", rendered)
 
         with open(self.paths["logs"], "a") as f:
-            writer = csv.DictWriter(f, fieldnames=self.fieldnames)
-            writer.writerow(to_log)
+            csv.DictWriter(f, fieldnames=self.fieldnames).writerow(to_log)
 
     def close(self, successful: bool = True) -> None:
         self.metadata["date_end"] = datetime.datetime.now().isoformat()

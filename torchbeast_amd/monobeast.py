@@ -143,7 +143,7 @@ def act(
 
         raw_env = create_env(flags)
         if hasattr(raw_env, "seed"):
-            raw_env.seed(actor_index ^ int.from_bytes(os.urandom(4), "little"))
+            raw_env.seed(actor_index)
         env = environment.Environment(raw_env)
 
         step_out = env.initial()
@@ -344,6 +344,16 @@ def train(flags):  # noqa: C901
         )
         p.start()
         actor_processes.append(p)
+
+    if flags.pin_buffers and flags.device.type == "cuda":
+        # Page-lock the already-shared rollout slots in this (learner)
+        # process so get_batch's H2D copies are true async DMA. hipHostRegister
+        # works on fork-shared mappings; the actors keep writing through their
+        # own (unregistered) mapping of the same pages.
+        cudart = torch.cuda.cudart()
+        for tensors in buffers.values():
+            for t in tensors:
+                cudart.cudaHostRegister(t.data_ptr(), t.numel() * t.element_size(), 0)
 
     learner_model = Net(obs_shape, num_actions, flags.use_lstm)
     learner_model = learner_model.to(device=flags.device)

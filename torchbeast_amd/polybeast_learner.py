@@ -377,6 +377,11 @@ def train(flags):  # noqa: C901
     with torch.no_grad():
         actor_flat.copy_(flat_param)
 
+    if getattr(flags, "momentum", 0):
+        raise ValueError(
+            "polybeast's fused RMSProp implements momentum=0 semantics only; "
+            "got --momentum %r (use monobeast for momentum>0)" % flags.momentum
+        )
     optimizer = tbflat.FusedRMSProp(
         flat_param,
         flat_grad,

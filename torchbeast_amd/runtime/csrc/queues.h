@@ -355,7 +355,12 @@ class DynamicBatcher {
       if (!fulfilled_) {
         auto eptr = std::make_exception_ptr(
             AsyncError("Batch destroyed before set_outputs was called"));
-        for (auto& req : requests_) req.promise->set_exception(eptr);
+        // fulfilled_count_ promises already carry values (set_outputs threw
+        // mid-way); breaking those again would raise future_error out of a
+        // destructor, so only fail the rest.
+        for (size_t i = fulfilled_count_; i < requests_.size(); ++i) {
+          requests_[i].promise->set_exception(eptr);
+        }
       }
     }
 
@@ -394,14 +399,23 @@ class DynamicBatcher {
           }
         });
       }
+      // Slice for every caller BEFORE fulfilling any promise: narrow() can
+      // throw on a mis-shaped output (when check_outputs is off), and a
+      // half-fulfilled batch would make the destructor re-break satisfied
+      // promises.
+      std::vector<TensorNest> slices;
+      slices.reserve(requests_.size());
       int64_t offset = 0;
       for (auto& req : requests_) {
-        TensorNest slice =
+        slices.push_back(
             outputs.map([this, offset, &req](const torch::Tensor& t) {
               return t.narrow(batch_dim_, offset, req.batch_size);
-            });
-        req.promise->set_value(std::move(slice));
+            }));
         offset += req.batch_size;
+      }
+      for (size_t i = 0; i < requests_.size(); ++i) {
+        requests_[i].promise->set_value(std::move(slices[i]));
+        fulfilled_count_ = i + 1;
       }
       fulfilled_ = true;
       if (service_stats_ != nullptr) {
@@ -416,6 +430,7 @@ class DynamicBatcher {
     StageStats* service_stats_;
     const int64_t created_us_;
     bool fulfilled_ = false;
+    size_t fulfilled_count_ = 0;
   };
 
   DynamicBatcher(int64_t batch_dim = 0,

@@ -146,7 +146,7 @@ class Reader {
   Nest<torch::Tensor> nest(int64_t prepend_ones) {
     uint8_t tag = u8();
     if (tag == kTagArray) {
-      // Rewind the tag for tensor() which expects to start at dtype.
+      // The array tag is consumed; tensor() parses from the dtype byte on.
       return Nest<torch::Tensor>(tensor(prepend_ones));
     }
     if (tag == kTagVector) {
