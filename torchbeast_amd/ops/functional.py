@@ -260,65 +260,99 @@ def rmsprop_step(param, grad, square_avg, lr, alpha, eps, clip_norm=None):
 # ---------------------------------------------------------------------------
 
 
-class _AtariTrunk(torch.autograd.Function):
+def _pack_trunk_weights(w1, w2, w3):
+    """bf16 operand layouts the MFMA trunk consumes: conv1 keeps the native
+    (c,ky,kx) flatten; conv2/3 are (ky,kx,c)-major for NHWC activations."""
+    w1p = w1.detach().reshape(w1.shape[0], -1).to(torch.bfloat16).contiguous()
+    w2p = (w2.detach().permute(0, 2, 3, 1).reshape(w2.shape[0], -1)
+           .to(torch.bfloat16).contiguous())
+    w3p = (w3.detach().permute(0, 2, 3, 1).reshape(w3.shape[0], -1)
+           .to(torch.bfloat16).contiguous())
+    return w1p, w2p, w3p
+
+
+class _AtariTrunkMfma(torch.autograd.Function):
+    """bf16 MFMA implicit-GEMM trunk (conv_mfma.hip): hand-written forward,
+    dgrad and wgrad kernels; fp32 accumulation throughout. Replaces the
+    reference's stock conv stack (torchbeast/monobeast.py:552-559,586-589)
+    on the learner path."""
+
     @staticmethod
     def forward(ctx, frames, w1, b1, w2, b2, w3, b3):
-        ext = _ext_for(frames, "atari_trunk_fwd")
-        out3, save1, save2 = ext.atari_trunk_fwd(
-            frames, w1, b1, w2, b2, w3, b3, True
+        ext = _ext_for(frames, "conv_trunk_fwd")
+        w1p, w2p, w3p = _pack_trunk_weights(w1, w2, w3)
+        out3, a1, a2 = ext.conv_trunk_fwd(
+            frames, w1p, b1.detach().contiguous(), w2p,
+            b2.detach().contiguous(), w3p, b3.detach().contiguous(), True
         )
-        ctx.save_for_backward(frames, w1, w2, w3, save1, save2, out3)
+        ctx.save_for_backward(frames, w2, w3, a1, a2, out3)
         return out3
 
     @staticmethod
     def backward(ctx, d_out3):
-        # Backward = relu masks + library convolution_backward per layer
-        # (GEMM-shaped reductions; see ops/hip/atari_trunk.hip for the
-        # hand-written variant kept for reference/benchmarks).
-        frames, w1, w2, w3, save1, save2, out3 = ctx.saved_tensors
-        N = frames.shape[0]
-        H2, W2 = save2.shape[2], save2.shape[3]
-        d3 = (d_out3 * (out3 > 0)).view(N, 64, H2 - 2, W2 - 2)
-
-        def conv_bwd(grad_out, inp, weight, stride, need_input_grad):
-            return torch.ops.aten.convolution_backward(
-                grad_out, inp, weight,
-                [weight.shape[0]],  # bias sizes
-                [stride, stride], [0, 0], [1, 1], False, [0, 0], 1,
-                [need_input_grad, True, True],
-            )
-
-        d2_pre, dw3, db3 = conv_bwd(d3, save2, w3, 1, True)
-        d2 = d2_pre * (save2 > 0)
-        d1_pre, dw2, db2 = conv_bwd(d2, save1, w2, 2, True)
-        d1 = d1_pre * (save1 > 0)
-        frames_f = frames.float().mul_(1.0 / 255.0)
-        _, dw1, db1 = conv_bwd(d1, frames_f, w1, 4, False)
+        frames, w2, w3, a1, a2, out3 = ctx.saved_tensors
+        ext = ops_mod.require_ext()
+        d3m = ext.conv_trunk_mask_d3(d_out3.contiguous().float(), out3)
+        # dgrad = stride-1 correlation with rotated weights, [ci][ky,kx,co].
+        w3r = (w3.detach().flip(2, 3).permute(1, 2, 3, 0)
+               .reshape(w3.shape[1], -1).to(torch.bfloat16).contiguous())
+        w2r = (w2.detach().flip(2, 3).permute(1, 2, 3, 0)
+               .reshape(w2.shape[1], -1).to(torch.bfloat16).contiguous())
+        d2 = ext.conv_trunk_dgrad3(d3m, w3r, a2)
+        d1 = ext.conv_trunk_dgrad2(d2, w2r, a1)
+        dw3p, db3 = ext.conv_trunk_wgrad3(a2, d3m)
+        dw2p, db2 = ext.conv_trunk_wgrad2(a1, d2)
+        dw1p, db1 = ext.conv_trunk_wgrad1(frames, d1)
+        # Kernel dW layout is [ky][co][kx*ci+c] ([ky][co][c*kw+kx] for conv1);
+        # permute back to PyTorch [co][ci][kh][kw].
+        dw3 = dw3p.view(3, 64, 3, 64).permute(1, 3, 0, 2).contiguous()
+        dw2 = dw2p.view(4, 64, 4, 32).permute(1, 3, 0, 2).contiguous()
+        dw1 = dw1p.view(8, 32, 4, 8).permute(1, 2, 0, 3).contiguous()
         return None, dw1, db1, dw2, db2, dw3, db3
 
 
 def atari_trunk(frames, conv1, conv2, conv3):
     """Fused u8-frame conv trunk for GPU [N,C,H,W] u8 frames.
 
-    Returns flat post-relu conv3 features [N, 64*H3*W3], or None when the
-    fused kernel doesn't apply (CPU tensors, or geometry exceeding LDS,
-    e.g. 210x160 full-res frames) — caller falls back to eager convs.
+    Dispatch (both paths are hand-written gfx950 kernels):
+    - training or large batches: bf16 MFMA implicit-GEMM kernels
+      (conv_mfma.hip) — covers the full learner batch T*B;
+    - small no-grad batches (inference service): the single-launch
+      per-sample fused trunk (atari_trunk.hip).
+    Returns flat post-relu conv3 features [N, 64*H3*W3] (NCHW flatten
+    order), or None when no kernel applies (CPU tensors, unsupported
+    geometry) — caller falls back to eager convs.
     """
     if not frames.is_cuda or frames.dtype != torch.uint8:
         return None
-    # Measured crossover vs MIOpen on MI355X (profiles/trunkbench): the
-    # per-sample fused kernel wins below ~384 samples (single launch, no f32
-    # frame materialization); larger learner batches use library convs.
-    if frames.shape[0] > 384 and not os.environ.get("TBAMD_FORCE_TRUNK"):
+    force = os.environ.get("TBAMD_TRUNK")  # mfma | valu | lib (benchmarking)
+    if force == "lib":
         return None
+    mfma_ok = tuple(frames.shape[1:]) == (4, 84, 84) and force != "valu"
+    grad = torch.is_grad_enabled() and (
+        conv1.weight.requires_grad or frames.requires_grad
+    )
+    if mfma_ok and (grad or frames.shape[0] > 384 or force == "mfma"):
+        ext = _ext_for(frames, "conv_trunk_fwd")
+        args = (frames.contiguous(), conv1.weight, conv1.bias, conv2.weight,
+                conv2.bias, conv3.weight, conv3.bias)
+        if grad:
+            return _AtariTrunkMfma.apply(*args)
+        w1p, w2p, w3p = _pack_trunk_weights(
+            conv1.weight, conv2.weight, conv3.weight)
+        (out3,) = ext.conv_trunk_fwd(
+            frames.contiguous(), w1p, conv1.bias.detach().contiguous(), w2p,
+            conv2.bias.detach().contiguous(), w3p,
+            conv3.bias.detach().contiguous(), False)
+        return out3
+    if grad:
+        return None  # unsupported geometry: eager autograd convs
     ext = _ext_for(frames, "atari_trunk_fwd")
     if ext is None or not ext.atari_trunk_supported(*frames.shape[1:]):
         return None
-    args = (frames.contiguous(), conv1.weight, conv1.bias, conv2.weight,
-            conv2.bias, conv3.weight, conv3.bias)
-    if torch.is_grad_enabled():
-        return _AtariTrunk.apply(*args)
-    (out3,) = ext.atari_trunk_fwd(*args, False)
+    (out3,) = ext.atari_trunk_fwd(
+        frames.contiguous(), conv1.weight, conv1.bias, conv2.weight,
+        conv2.bias, conv3.weight, conv3.bias, False)
     return out3
 
 

@@ -26,10 +26,6 @@ std::vector<torch::Tensor> atari_trunk_fwd(
 
 // d_out3: [N, 64*H3*W3] upstream gradient (pre relu-mask).
 // Returns {dw1, db1, dw2, db2, dw3, db3}.
-std::vector<torch::Tensor> atari_trunk_bwd(
-    torch::Tensor frames, torch::Tensor w1, torch::Tensor w2,
-    torch::Tensor w3, torch::Tensor out1, torch::Tensor out2,
-    torch::Tensor out3_flat, torch::Tensor d_out3);
 
 // bf16 MFMA GEMM (mfma_gemm.hip): C[M,N] f32 = A[M,K] @ B[N,K]^T.
 torch::Tensor mfma_gemm(torch::Tensor A, torch::Tensor B);

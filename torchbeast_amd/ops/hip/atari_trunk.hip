@@ -194,134 +194,6 @@ __global__ __launch_bounds__(kFwdThreads) void trunk_fwd_kernel(
   }
 }
 
-// Reverse chain per sample: d3 = d_out3 * relu'(out3); d2 = conv3^T(d3)
-// * relu'(out2); d1 = conv2^T(d2) * relu'(out1). d3/d2/d1 are written to
-// global for the wgrad kernels. LDS: d3 + d2 planes.
-__global__ __launch_bounds__(kThreads) void trunk_bwd_dgrad_kernel(
-    const float* __restrict__ d_out3, const float* __restrict__ out3,
-    const float* __restrict__ out2, const float* __restrict__ out1,
-    const float* __restrict__ w3, const float* __restrict__ w2, int H1,
-    int W1, int H2, int W2, int H3, int W3, float* __restrict__ g3,
-    float* __restrict__ g2, float* __restrict__ g1) {
-  extern __shared__ unsigned char smem[];
-  float* s_d3 = reinterpret_cast<float*>(smem);          // 64*H3*W3
-  float* s_d2 = s_d3 + 64 * H3 * W3;                     // 64*H2*W2
-
-  const int n = blockIdx.x;
-  const int tid = threadIdx.x;
-  const int n3 = 64 * H3 * W3;
-  const int n2 = 64 * H2 * W2;
-  const int n1 = 32 * H1 * W1;
-
-  for (int i = tid; i < n3; i += kThreads) {
-    const float v = out3[(int64_t)n * n3 + i] > 0.f
-                        ? d_out3[(int64_t)n * n3 + i]
-                        : 0.f;
-    s_d3[i] = v;
-    g3[(int64_t)n * n3 + i] = v;
-  }
-  __syncthreads();
-
-  {  // conv3 dgrad (k3 s1): d2[ci,y,x] = sum_{co,ky,kx} d3[co,y-ky,x-kx]*w3.
-    const int plane2 = H2 * W2;
-    for (int i = tid; i < n2; i += kThreads) {
-      const int ci = i / plane2;
-      const int r = i - ci * plane2;
-      const int y = r / W2, x = r - (r / W2) * W2;
-      float acc = 0.f;
-      for (int co = 0; co < 64; ++co) {
-        const float* d3_c = s_d3 + co * H3 * W3;
-        const float* w_c = w3 + ((int64_t)co * 64 + ci) * 9;
-#pragma unroll
-        for (int ky = 0; ky < 3; ++ky) {
-          const int oy = y - ky;
-          if (oy < 0 || oy >= H3) continue;
-#pragma unroll
-          for (int kx = 0; kx < 3; ++kx) {
-            const int ox = x - kx;
-            if (ox < 0 || ox >= W3) continue;
-            acc += d3_c[oy * W3 + ox] * w_c[ky * 3 + kx];
-          }
-        }
-      }
-      const float v = out2[(int64_t)n * n2 + i] > 0.f ? acc : 0.f;
-      s_d2[i] = v;
-      g2[(int64_t)n * n2 + i] = v;
-    }
-  }
-  __syncthreads();
-
-  {  // conv2 dgrad (k4 s2): d1[ci,y,x] over taps with (y-ky)%2==0.
-    const int plane1 = H1 * W1;
-    for (int i = tid; i < n1; i += kThreads) {
-      const int ci = i / plane1;
-      const int r = i - ci * plane1;
-      const int y = r / W1, x = r - (r / W1) * W1;
-      float acc = 0.f;
-      for (int co = 0; co < 64; ++co) {
-        const float* d2_c = s_d2 + co * H2 * W2;
-        const float* w_c = w2 + ((int64_t)co * 32 + ci) * 16;
-#pragma unroll
-        for (int ky = 0; ky < 4; ++ky) {
-          const int ty = y - ky;
-          if (ty < 0 || (ty & 1) || ty / 2 >= H2) continue;
-#pragma unroll
-          for (int kx = 0; kx < 4; ++kx) {
-            const int tx = x - kx;
-            if (tx < 0 || (tx & 1) || tx / 2 >= W2) continue;
-            acc += d2_c[(ty / 2) * W2 + tx / 2] * w_c[ky * 4 + kx];
-          }
-        }
-      }
-      const float v = out1[(int64_t)n * n1 + i] > 0.f ? acc : 0.f;
-      g1[(int64_t)n * n1 + i] = v;
-    }
-  }
-}
-
-// dw[co,ci,ky,kx] = sum_n sum_{oy,ox} d[n,co,oy,ox] * x[n,ci,oy*s+ky,ox*s+kx]
-// One block per (co,ky,kx); threads split as (ci, n-group); LDS-reduced.
-// IN_T is float (activations) or uint8_t (frames, with 1/255 fold).
-template <typename IN_T>
-__global__ __launch_bounds__(kThreads) void conv_wgrad_kernel(
-    const float* __restrict__ d, const IN_T* __restrict__ x, int N, int Cin,
-    int Hx, int Wx, int Ho, int Wo, int K, int stride, float scale,
-    float* __restrict__ dw) {
-  __shared__ float red[kThreads];
-  const int b = blockIdx.x;  // co*K*K + ky*K + kx
-  const int co = b / (K * K);
-  const int ky = (b / K) % K;
-  const int kx = b % K;
-
-  const int groups = kThreads / Cin;  // Cin in {4, 32, 64} divides 256.
-  const int ci = threadIdx.x % Cin;
-  const int grp = threadIdx.x / Cin;
-
-  float acc = 0.f;
-  if (grp < groups) {
-    const int64_t Cout_plane = (int64_t)Ho * Wo;
-    for (int n = grp; n < N; n += groups) {
-      const float* d_nc = d + ((int64_t)n * gridDim.x / (K * K) + co) * Cout_plane;
-      const IN_T* x_nc = x + ((int64_t)n * Cin + ci) * Hx * Wx + ky * Wx + kx;
-      for (int oy = 0; oy < Ho; ++oy) {
-        const float* d_row = d_nc + oy * Wo;
-        const IN_T* x_row = x_nc + (oy * stride) * Wx;
-        for (int ox = 0; ox < Wo; ++ox) {
-          acc += d_row[ox] * (float)x_row[ox * stride];
-        }
-      }
-    }
-  }
-  red[threadIdx.x] = acc;
-  __syncthreads();
-  // Reduce over n-groups for each ci: entries ci, ci+Cin, ci+2*Cin, ...
-  if (threadIdx.x < Cin) {
-    float sum = 0.f;
-    for (int g = 0; g < groups; ++g) sum += red[g * Cin + threadIdx.x];
-    dw[((int64_t)co * Cin + threadIdx.x) * K * K + ky * K + kx] = sum * scale;
-  }
-}
-
 }  // namespace
 
 bool atari_trunk_supported(int64_t C, int64_t H, int64_t W) {
@@ -373,54 +245,6 @@ std::vector<torch::Tensor> atari_trunk_fwd(
                      save1_p, save2_p);
   if (save_for_backward) return {out3, save1, save2};
   return {out3};
-}
-
-std::vector<torch::Tensor> atari_trunk_bwd(
-    torch::Tensor frames, torch::Tensor w1, torch::Tensor w2,
-    torch::Tensor w3, torch::Tensor out1, torch::Tensor out2,
-    torch::Tensor out3_flat, torch::Tensor d_out3) {
-  auto fr = frames.contiguous();
-  const int64_t N = fr.size(0);
-  Geom g = make_geom(fr.size(1), fr.size(2), fr.size(3));
-  auto opts = w1.options();
-  auto stream = at::cuda::getCurrentCUDAStream();
-
-  auto g3 = torch::empty({N, 64, g.H3, g.W3}, opts);
-  auto g2 = torch::empty({N, 64, g.H2, g.W2}, opts);
-  auto g1 = torch::empty({N, 32, g.H1, g.W1}, opts);
-
-  const size_t lds = ((size_t)64 * g.H3 * g.W3 + (size_t)64 * g.H2 * g.W2) * 4;
-  hipLaunchKernelGGL(trunk_bwd_dgrad_kernel, dim3(N), dim3(kThreads), lds,
-                     stream, d_out3.contiguous().data_ptr<float>(),
-                     out3_flat.data_ptr<float>(), out2.data_ptr<float>(),
-                     out1.data_ptr<float>(), w3.contiguous().data_ptr<float>(),
-                     w2.contiguous().data_ptr<float>(), g.H1, g.W1, g.H2,
-                     g.W2, g.H3, g.W3, g3.data_ptr<float>(),
-                     g2.data_ptr<float>(), g1.data_ptr<float>());
-
-  auto dw3 = torch::empty_like(w3);
-  auto dw2 = torch::empty_like(w2);
-  auto dw1 = torch::empty_like(w1);
-  // conv3 wgrad: d=g3, x=out2.
-  hipLaunchKernelGGL(conv_wgrad_kernel<float>, dim3(64 * 9), dim3(kThreads),
-                     0, stream, g3.data_ptr<float>(), out2.data_ptr<float>(),
-                     N, 64, g.H2, g.W2, g.H3, g.W3, 3, 1, 1.f,
-                     dw3.data_ptr<float>());
-  // conv2 wgrad: d=g2, x=out1.
-  hipLaunchKernelGGL(conv_wgrad_kernel<float>, dim3(64 * 16), dim3(kThreads),
-                     0, stream, g2.data_ptr<float>(), out1.data_ptr<float>(),
-                     N, 32, g.H1, g.W1, g.H2, g.W2, 4, 2, 1.f,
-                     dw2.data_ptr<float>());
-  // conv1 wgrad: d=g1, x=frames u8 (fold the 1/255 normalize).
-  hipLaunchKernelGGL(conv_wgrad_kernel<uint8_t>, dim3(32 * 64), dim3(kThreads),
-                     0, stream, g1.data_ptr<float>(), fr.data_ptr<uint8_t>(),
-                     N, g.C, g.H, g.W, g.H1, g.W1, 8, 4, 1.f / 255.f,
-                     dw1.data_ptr<float>());
-
-  auto db3 = g3.sum({0, 2, 3});
-  auto db2 = g2.sum({0, 2, 3});
-  auto db1 = g1.sum({0, 2, 3});
-  return {dw1, db1, dw2, db2, dw3, db3};
 }
 
 }  // namespace tbamd

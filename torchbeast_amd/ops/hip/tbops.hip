@@ -26,6 +26,7 @@
 #include <vector>
 
 #include "atari_trunk.h"
+#include "conv_mfma.h"
 
 namespace cg = cooperative_groups;
 
@@ -746,8 +747,14 @@ std::vector<torch::Tensor> lstm_unroll_bwd(
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("atari_trunk_fwd", &tbamd::atari_trunk_fwd);
-  m.def("atari_trunk_bwd", &tbamd::atari_trunk_bwd);
   m.def("atari_trunk_supported", &tbamd::atari_trunk_supported);
+  m.def("conv_trunk_fwd", &tbamd::conv_trunk_fwd);
+  m.def("conv_trunk_mask_d3", &tbamd::conv_trunk_mask_d3);
+  m.def("conv_trunk_dgrad3", &tbamd::conv_trunk_dgrad3);
+  m.def("conv_trunk_dgrad2", &tbamd::conv_trunk_dgrad2);
+  m.def("conv_trunk_wgrad1", &tbamd::conv_trunk_wgrad1);
+  m.def("conv_trunk_wgrad2", &tbamd::conv_trunk_wgrad2);
+  m.def("conv_trunk_wgrad3", &tbamd::conv_trunk_wgrad3);
   m.def("mfma_gemm", &tbamd::mfma_gemm);
   m.def("mfma_gemm_probe", &tbamd::mfma_gemm_probe);
   m.def("mfma_gemm_v2", &tbamd::mfma_gemm_v2);
