@@ -154,7 +154,9 @@ def main():
     )
     total_env_steps = (args.warmup + args.steps) * T * B * world_size * 10
     scheduler = tbflat.LinearLR(optimizer, T * B * world_size, total_env_steps)
-    reducer = tbddp.GradAllReducer(flat_grad, world_size)
+    reducer = tbddp.GradAllReducer(
+        flat_grad, world_size,
+        params=[p for p in model.parameters() if p.requires_grad])
 
     learner_queue = runtime.BatchingQueue(
         batch_dim=1, minimum_batch_size=B, maximum_batch_size=B,

@@ -28,6 +28,11 @@
 #include "atari_trunk.h"
 #include "conv_mfma.h"
 
+namespace tbamd {
+std::vector<double> barrier_probe(int64_t variant, int64_t wgs, int64_t iters);
+double launch_probe(int64_t iters);
+}  // namespace tbamd (probes.hip)
+
 namespace cg = cooperative_groups;
 
 #define DEVCHECK(x) TORCH_CHECK(x == hipSuccess, "HIP error: ", hipGetErrorString(x))
@@ -755,6 +760,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_trunk_wgrad1", &tbamd::conv_trunk_wgrad1);
   m.def("conv_trunk_wgrad2", &tbamd::conv_trunk_wgrad2);
   m.def("conv_trunk_wgrad3", &tbamd::conv_trunk_wgrad3);
+  m.def("barrier_probe", &tbamd::barrier_probe);
+  m.def("launch_probe", &tbamd::launch_probe);
   m.def("mfma_gemm", &tbamd::mfma_gemm);
   m.def("mfma_gemm_probe", &tbamd::mfma_gemm_probe);
   m.def("mfma_gemm_v2", &tbamd::mfma_gemm_v2);

@@ -41,20 +41,110 @@ def maybe_init_distributed():
 
 
 class GradAllReducer:
-    """All-reduce a flat gradient buffer across the data-parallel group."""
+    """All-reduce a flat gradient buffer across the data-parallel group.
 
-    def __init__(self, flat_grad: torch.Tensor, world_size: int):
+    With `params` (the module's parameters in registration order, matching
+    the flat layout), gradients are reduced in BUCKETS launched from
+    post-accumulate-grad hooks: as soon as the tail of the model (heads,
+    fc) finishes its backward, its bucket's all-reduce starts on a side
+    stream and overlaps the conv trunk's backward. Without `params`, one
+    all-reduce of the whole buffer after backward (the ~20 MB AtariNet
+    payload is still a single fused call per bucket — sized for 7-link
+    point-to-point xGMI, not NVSwitch).
+    """
+
+    def __init__(self, flat_grad: torch.Tensor, world_size: int,
+                 params=None, num_buckets: int = 2):
         self.flat_grad = flat_grad
         self.world_size = world_size
         self.enabled = world_size > 1 and dist.is_initialized()
         self._stream = (
             torch.cuda.Stream() if self.enabled and flat_grad.is_cuda else None
         )
+        self._buckets = []       # list of (start, end) flat slices
+        self._pending = []       # per-bucket count of params not yet ready
+        self._launched = []      # per-bucket bool
+        self._handles = []
+        self._param_bucket = {}
+        if self.enabled and params is not None:
+            self._build_buckets(list(params), num_buckets)
+
+    def _build_buckets(self, params, num_buckets):
+        # Flat offsets follow registration order; backward completes roughly
+        # in reverse, so buckets are contiguous ranges assembled from the
+        # END of the flat buffer backwards.
+        offsets = []
+        off = 0
+        for p in params:
+            offsets.append((p, off, off + p.numel()))
+            off += p.numel()
+        total = off
+        target = (total + num_buckets - 1) // num_buckets
+        bucket_of = {}
+        bid, acc = 0, 0
+        for p, start, end in reversed(offsets):
+            if acc >= target and bid + 1 < num_buckets:
+                bid, acc = bid + 1, 0
+            bucket_of[p] = bid
+            acc += end - start
+        nb = bid + 1
+        spans = [[None, None] for _ in range(nb)]
+        counts = [0] * nb
+        for p, start, end in offsets:
+            b = bucket_of[p]
+            counts[b] += 1
+            if spans[b][0] is None or start < spans[b][0]:
+                spans[b][0] = start
+            if spans[b][1] is None or end > spans[b][1]:
+                spans[b][1] = end
+        self._buckets = [tuple(s) for s in spans]
+        self._counts = counts
+        self._pending = list(counts)
+        self._launched = [False] * nb
+        for p, _, _ in offsets:
+            b = bucket_of[p]
+            self._param_bucket[p] = b
+            p.register_post_accumulate_grad_hook(self._make_hook(b))
+
+    def _make_hook(self, bucket_id):
+        def hook(_param):
+            self._pending[bucket_id] -= 1
+            if self._pending[bucket_id] == 0:
+                self._launch(bucket_id)
+        return hook
+
+    def _launch(self, bucket_id):
+        if self._launched[bucket_id]:
+            return
+        self._launched[bucket_id] = True
+        start, end = self._buckets[bucket_id]
+        chunk = self.flat_grad[start:end]
+        if self._stream is not None:
+            self._stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._stream):
+                chunk.div_(self.world_size)
+                self._handles.append(dist.all_reduce(chunk, async_op=True))
+        else:
+            chunk.div_(self.world_size)
+            self._handles.append(dist.all_reduce(chunk, async_op=True))
 
     def reduce(self):
-        """Average gradients across ranks. Call between backward() and
-        optimizer.step()."""
+        """Finish averaging gradients across ranks. Call between backward()
+        and optimizer.step(); bucketed reduces launched from backward hooks
+        are awaited here, anything not yet launched is launched now."""
         if not self.enabled:
+            return
+        if self._buckets:
+            for b in range(len(self._buckets)):
+                if not self._launched[b]:
+                    self._launch(b)
+            for h in self._handles:
+                h.wait()
+            self._handles.clear()
+            if self._stream is not None:
+                torch.cuda.current_stream().wait_stream(self._stream)
+            self._pending = list(self._counts)
+            self._launched = [False] * len(self._buckets)
             return
         self.flat_grad.div_(self.world_size)
         if self._stream is not None:

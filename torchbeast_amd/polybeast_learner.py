@@ -286,7 +286,9 @@ def learn(
                 actor_flat.copy_(flat_param)
 
             episode_returns = env_outputs.episode_return[env_outputs.done]
-            stats["step"] = stats.get("step", 0) + flags.unroll_length * flags.batch_size
+            stats["step"] = (stats.get("step", 0)
+                         + flags.unroll_length * flags.batch_size
+                         * max(1, reducer.world_size))
             stats["episode_returns"] = tuple(episode_returns.cpu().numpy())
             stats["mean_episode_return"] = torch.mean(episode_returns).item()
             stats["mean_episode_step"] = torch.mean(
@@ -392,7 +394,9 @@ def train(flags):  # noqa: C901
     )
     steps_per_update = flags.unroll_length * flags.batch_size * world_size
     scheduler = tbflat.LinearLR(optimizer, steps_per_update, flags.total_steps)
-    reducer = tbddp.GradAllReducer(flat_grad, world_size)
+    reducer = tbddp.GradAllReducer(
+        flat_grad, world_size,
+        params=[p for p in model.parameters() if p.requires_grad])
 
     stats = {}
 
@@ -532,6 +536,7 @@ def train(flags):  # noqa: C901
             plogger.close()
 
     logging.info("Rank %d done after %d updates.", rank, update_counter["done"])
+    return model
 
 
 def test(flags, num_episodes: int = 10):
