@@ -31,6 +31,7 @@ ext_modules = [
         sources=[
             "torchbeast_amd/runtime/csrc/module.cc",
             "torchbeast_amd/ops/hip/atari_trunk.hip",
+            "torchbeast_amd/ops/hip/conv_mfma.hip",
         ],
         extra_compile_args={
             "cxx": ["-O3", "-std=c++17", "-pthread"] + _san_flags,

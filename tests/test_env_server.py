@@ -226,3 +226,36 @@ def test_wire_roundtrip_dtypes(dtype):
     expected0 = np.arange(6, dtype=dtype).reshape(2, 3)
     np.testing.assert_array_equal(frames[0, 0].numpy(), expected0)
     np.testing.assert_array_equal(frames[1, 0].numpy(), expected0 + 1)
+
+
+def test_tcp_env_roundtrip():
+    """Cross-machine-capable env plane: the same framed protocol over TCP
+    (the reference's gRPC plane worked over any channel; wire.h + tcp:
+    addresses restore that capability)."""
+    import socket as pysocket
+
+    s = pysocket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+
+    address = f"tcp:127.0.0.1:{port}"
+    server = runtime.Server(lambda: CountingEnv(episode_length=5), address)
+    server.start()
+    try:
+        def inference(env_outputs, agent_state):
+            frame = env_outputs[0]
+            b = frame.shape[1]
+            action = torch.zeros((1, b), dtype=torch.int64)
+            logits = torch.ones((1, b, 2))
+            baseline = torch.zeros((1, b))
+            return ((action, logits, baseline), agent_state)
+
+        rollouts = _run_pool(unroll_length=6, addresses=[address],
+                             inference_fn=inference, n_rollouts=2)
+        (env_outputs, _), _ = rollouts[0]
+        frames = env_outputs[0]
+        # CountingEnv counts 0..limit; rollout frames must follow it.
+        assert frames.shape[0] == 7  # T+1
+    finally:
+        server.stop()

@@ -8,6 +8,8 @@
 
 #pragma once
 
+#include <arpa/inet.h>
+#include <netinet/in.h>
 #include <pybind11/pybind11.h>
 #include <sys/socket.h>
 #include <sys/un.h>
@@ -29,7 +31,7 @@ namespace py = pybind11;
 class EnvServer {
  public:
   EnvServer(py::object env_init, std::string address)
-      : env_init_(std::move(env_init)),
+      : env_init_(std::move(env_init)), address_(address),
         path_(strip_unix_prefix(address)) {}
 
   ~EnvServer() {
@@ -41,16 +43,37 @@ class EnvServer {
 
   void start() {
     if (listen_fd_ >= 0) throw std::runtime_error("server already running");
-    ::unlink(path_.c_str());
-    listen_fd_ = ::socket(AF_UNIX, SOCK_STREAM, 0);
-    if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
-    sockaddr_un addr;
-    std::memset(&addr, 0, sizeof(addr));
-    addr.sun_family = AF_UNIX;
-    std::strncpy(addr.sun_path, path_.c_str(), sizeof(addr.sun_path) - 1);
-    if (::bind(listen_fd_, reinterpret_cast<sockaddr*>(&addr), sizeof(addr)) !=
-        0) {
-      throw std::runtime_error("bind(" + path_ + ") failed");
+    std::string host;
+    uint16_t port;
+    if (parse_tcp_address(address_, &host, &port)) {
+      // Cross-machine env plane: listen on TCP (the reference's gRPC server
+      // had this for free; here it is the same framed protocol over TCP).
+      listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
+      if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
+      int one = 1;
+      ::setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+      sockaddr_in addr;
+      std::memset(&addr, 0, sizeof(addr));
+      addr.sin_family = AF_INET;
+      addr.sin_port = htons(port);
+      addr.sin_addr.s_addr =
+          host.empty() ? INADDR_ANY : ::inet_addr(host.c_str());
+      if (::bind(listen_fd_, reinterpret_cast<sockaddr*>(&addr),
+                 sizeof(addr)) != 0) {
+        throw std::runtime_error("bind(" + address_ + ") failed");
+      }
+    } else {
+      ::unlink(path_.c_str());
+      listen_fd_ = ::socket(AF_UNIX, SOCK_STREAM, 0);
+      if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
+      sockaddr_un addr;
+      std::memset(&addr, 0, sizeof(addr));
+      addr.sun_family = AF_UNIX;
+      std::strncpy(addr.sun_path, path_.c_str(), sizeof(addr.sun_path) - 1);
+      if (::bind(listen_fd_, reinterpret_cast<sockaddr*>(&addr),
+                 sizeof(addr)) != 0) {
+        throw std::runtime_error("bind(" + path_ + ") failed");
+      }
     }
     if (::listen(listen_fd_, 128) != 0) {
       throw std::runtime_error("listen() failed");
@@ -198,6 +221,7 @@ class EnvServer {
   }
 
   py::object env_init_;
+  std::string address_;
   std::string path_;
   int listen_fd_ = -1;
   std::atomic<bool> running_{false};

@@ -16,6 +16,9 @@
 
 #pragma once
 
+#include <netdb.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
 #include <sys/socket.h>
 #include <sys/un.h>
 #include <unistd.h>
@@ -111,6 +114,56 @@ inline std::string strip_unix_prefix(const std::string& address) {
   return address;
 }
 
+// "tcp:host:port" -> (host, port); returns false for non-TCP addresses.
+// The reference's gRPC env plane worked over any channel
+// (ref: src/proto/rpcenv.proto:46-48); this restores the cross-machine
+// capability for the framed wire protocol.
+inline bool parse_tcp_address(const std::string& address, std::string* host,
+                              uint16_t* port) {
+  if (address.rfind("tcp:", 0) != 0) return false;
+  const std::string rest = address.substr(4);
+  const size_t colon = rest.rfind(':');
+  if (colon == std::string::npos) {
+    throw std::runtime_error("tcp address must be tcp:host:port, got " +
+                             address);
+  }
+  *host = rest.substr(0, colon);
+  *port = static_cast<uint16_t>(std::stoi(rest.substr(colon + 1)));
+  return true;
+}
+
+inline int connect_tcp(const std::string& host, uint16_t port,
+                       std::chrono::seconds deadline) {
+  addrinfo hints;
+  std::memset(&hints, 0, sizeof(hints));
+  hints.ai_family = AF_UNSPEC;
+  hints.ai_socktype = SOCK_STREAM;
+  const std::string port_s = std::to_string(port);
+  auto start = std::chrono::steady_clock::now();
+  for (;;) {
+    addrinfo* res = nullptr;
+    if (::getaddrinfo(host.c_str(), port_s.c_str(), &hints, &res) == 0) {
+      for (addrinfo* ai = res; ai != nullptr; ai = ai->ai_next) {
+        int fd = ::socket(ai->ai_family, ai->ai_socktype, ai->ai_protocol);
+        if (fd < 0) continue;
+        if (::connect(fd, ai->ai_addr, ai->ai_addrlen) == 0) {
+          int one = 1;
+          ::setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+          ::freeaddrinfo(res);
+          return fd;
+        }
+        ::close(fd);
+      }
+      ::freeaddrinfo(res);
+    }
+    if (std::chrono::steady_clock::now() - start > deadline) {
+      throw std::runtime_error("timeout connecting to env server at tcp:" +
+                               host + ":" + port_s);
+    }
+    std::this_thread::sleep_for(std::chrono::milliseconds(50));
+  }
+}
+
 inline int connect_unix(const std::string& address,
                         std::chrono::seconds deadline) {
   std::string path = strip_unix_prefix(address);
@@ -144,10 +197,20 @@ class EnvConnection {
   virtual TensorNest step(const torch::Tensor& action) = 0;
 };
 
+inline int connect_env_socket(const std::string& address,
+                              std::chrono::seconds deadline) {
+  std::string host;
+  uint16_t port;
+  if (parse_tcp_address(address, &host, &port)) {
+    return connect_tcp(host, port, deadline);
+  }
+  return connect_unix(address, deadline);
+}
+
 class SocketEnv : public EnvConnection {
  public:
   SocketEnv(const std::string& address, std::chrono::seconds connect_deadline)
-      : stream_(connect_unix(address, connect_deadline)) {}
+      : stream_(connect_env_socket(address, connect_deadline)) {}
 
   TensorNest initial() override { return read_step(); }
 

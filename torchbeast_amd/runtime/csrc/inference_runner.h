@@ -23,6 +23,7 @@
 #include <vector>
 
 #include "../../ops/hip/atari_trunk.h"
+#include "../../ops/hip/conv_mfma.h"
 #include "queues.h"
 
 namespace tbruntime {
@@ -168,9 +169,31 @@ class InferenceRunner {
         }
       }
       x = at::relu(x).reshape({bp, -1});
+    } else if (C == 4 && H == 84 && W == 84) {
+      // MFMA implicit-GEMM trunk (bf16 operands, fp32 accumulate) — flat
+      // per-batch cost at every dynamic batch size, unlike the per-sample
+      // fused kernel whose grid is the batch (underfills the 256 CUs below
+      // ~512 samples and was measured dominating GPU time at small
+      // batches: profiles/PROFILE_r2.md).
+      x = tbamd::conv_trunk_fwd(
+          frames_p,
+          weights_[0].reshape({32, -1}).to(torch::kBFloat16).contiguous(),
+          weights_[1],
+          weights_[2]
+              .permute({0, 2, 3, 1})
+              .reshape({64, -1})
+              .to(torch::kBFloat16)
+              .contiguous(),
+          weights_[3],
+          weights_[4]
+              .permute({0, 2, 3, 1})
+              .reshape({64, -1})
+              .to(torch::kBFloat16)
+              .contiguous(),
+          weights_[5], /*want_stash=*/false)[0];
     } else if (bp <= 384 && tbamd::atari_trunk_supported(C, H, W)) {
       // Hand-written fused CDNA4 conv trunk: one kernel for the u8
-      // normalize + 3 convs.
+      // normalize + 3 convs (non-84x84 geometries).
       x = tbamd::atari_trunk_fwd(frames_p, weights_[0], weights_[1],
                                  weights_[2], weights_[3], weights_[4],
                                  weights_[5], /*save_for_backward=*/false)[0];
