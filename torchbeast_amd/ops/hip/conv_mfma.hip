@@ -218,8 +218,8 @@ __global__ __launch_bounds__(kThreads) void conv1_u8_kernel(
 // ---------------------------------------------------------------------------
 
 template <int CI, int IHR, int IWR, int PAD, int DIL, int KH, int KW, int ST,
-          int CO, int OH, int OW, int SB, int OYT, int MODE>
-__global__ __launch_bounds__(kThreads) void conv_nhwc_kernel(
+          int CO, int OH, int OW, int SB, int OYT, int MODE, int NT = kThreads>
+__global__ __launch_bounds__(NT) void conv_nhwc_kernel(
     const __bf16* __restrict__ in,   // [N, IHR, IWR, CI]
     const __bf16* __restrict__ W,    // [CO, KH*KW*CI] (ky,kx,c)-major
     const float* __restrict__ bias,  // [CO] (MODE 0/1)
@@ -233,7 +233,8 @@ __global__ __launch_bounds__(kThreads) void conv_nhwc_kernel(
   constexpr int M_BLK = SB * OYT * OW;
   constexpr int MF = (M_BLK + 15) / 16;
   constexpr int NF = CO / 16;
-  constexpr int MAX_MF = (MF + kWaves - 1) / kWaves;
+  constexpr int NW = NT / 64;
+  constexpr int MAX_MF = (MF + NW - 1) / NW;
   constexpr int BANDS = (OH + OYT - 1) / OYT;
   static_assert(ROWE % 8 == 0, "LDS rows must be whole 16 B chunks");
   static_assert((KW * CI) % 32 == 0, "K-steps must not cross ky rows");
@@ -258,7 +259,7 @@ __global__ __launch_bounds__(kThreads) void conv_nhwc_kernel(
   // ---- stage the logical input tile (zeros for pad/dilation gaps) ----
   {
     constexpr int NCHUNK = SB * LROWS * ROWE / 8;
-    for (int idx = tid; idx < NCHUNK; idx += kThreads) {
+    for (int idx = tid; idx < NCHUNK; idx += NT) {
       const int e0 = idx * 8;
       const int sl = e0 / (LROWS * ROWE);
       const int rem = e0 % (LROWS * ROWE);
@@ -292,7 +293,7 @@ __global__ __launch_bounds__(kThreads) void conv_nhwc_kernel(
 
   int aoff[MAX_MF];
   int nmf = 0;
-  for (int f = wave; f < MF; f += kWaves, ++nmf) {
+  for (int f = wave; f < MF; f += NW, ++nmf) {
     const int m = min(f * 16 + ln, M_BLK - 1);
     const int sl = m / (OYT * OW);
     const int rm = m % (OYT * OW);
@@ -332,7 +333,7 @@ __global__ __launch_bounds__(kThreads) void conv_nhwc_kernel(
 
   // ---- epilogue ----
   int fi = 0;
-  for (int f = wave; f < MF; f += kWaves, ++fi) {
+  for (int f = wave; f < MF; f += NW, ++fi) {
 #pragma unroll
     for (int j = 0; j < NF; ++j) {
       const int cch = j * 16 + ln;
@@ -621,6 +622,30 @@ std::vector<torch::Tensor> conv_trunk_fwd(torch::Tensor frames,
       dim3(N), dim3(kThreads), G1::LDS, stream, fr.data_ptr<uint8_t>(),
       reinterpret_cast<const __bf16*>(w1.data_ptr()), b1.data_ptr<float>(),
       reinterpret_cast<__bf16*>(a1.data_ptr()), N);
+
+  // Small (inference-sized) batches need grid = N for occupancy: the
+  // sample-grouped variants would launch only N/SB blocks (measured: conv3
+  // at batch ~100 was the top GPU kernel purely from underfill).
+  if (N < 1024) {
+    constexpr size_t kLds2 = (size_t)G2::IH * G2::IW * G2::CI * 2;
+    hipLaunchKernelGGL(
+        (conv_nhwc_kernel<G2::CI, G2::IH, G2::IW, 0, 1, G2::KH, G2::KW,
+                          G2::ST, G2::CO, G2::OH, G2::OW, 1, G2::OH, 0, 384>),
+        dim3(N), dim3(384), kLds2, stream,
+        reinterpret_cast<const __bf16*>(a1.data_ptr()),
+        reinterpret_cast<const __bf16*>(w2.data_ptr()), b2.data_ptr<float>(),
+        nullptr, a2.data_ptr(), N);
+    constexpr size_t kLds3 = (size_t)G3::IH * G3::IW * G3::CI * 2;
+    hipLaunchKernelGGL(
+        (conv_nhwc_kernel<G3::CI, G3::IH, G3::IW, 0, 1, G3::KH, G3::KW,
+                          G3::ST, G3::CO, G3::OH, G3::OW, 1, G3::OH, 1, 256>),
+        dim3(N), dim3(256), kLds3, stream,
+        reinterpret_cast<const __bf16*>(a2.data_ptr()),
+        reinterpret_cast<const __bf16*>(w3.data_ptr()), b3.data_ptr<float>(),
+        nullptr, out3.data_ptr(), N);
+    return want_stash ? std::vector<torch::Tensor>{out3, a1, a2}
+                      : std::vector<torch::Tensor>{out3};
+  }
 
   hipLaunchKernelGGL(
       (conv_nhwc_kernel<G2::CI, G2::IH, G2::IW, 0, 1, G2::KH, G2::KW, G2::ST,
