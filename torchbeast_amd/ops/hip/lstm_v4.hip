@@ -43,6 +43,8 @@ namespace {
 
 constexpr int kNT = 512;  // 8 waves
 
+using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
+
 __device__ __forceinline__ void agent_barrier(int* count, int* sense,
                                               int nblocks, int* lsense) {
   __syncthreads();
@@ -96,10 +98,11 @@ __global__ __launch_bounds__(kNT) void lstm4_fwd_kernel(
     float* __restrict__ cm,     // [T,B,H] masked c_{t-1} (stash)
     float* __restrict__ c_out,  // [T,B,H] (stash)
     float* __restrict__ hT, float* __restrict__ cT) {  // [B,H]
+  const int HP = (H + 7) & ~7;  // LDS rows padded to 16 B for b128 reads
   extern __shared__ char smem[];
-  __bf16* sW = reinterpret_cast<__bf16*>(smem);               // [4*hs][H]
-  __bf16* s_hm = sW + (size_t)4 * hs * H;                     // [B][H]
-  float* s_gates = reinterpret_cast<float*>(s_hm + (size_t)B * H);  // [4*hs*B]
+  __bf16* sW = reinterpret_cast<__bf16*>(smem);               // [4*hs][HP]
+  __bf16* s_hm = sW + (size_t)4 * hs * HP;                    // [B][HP]
+  float* s_gates = reinterpret_cast<float*>(s_hm + (size_t)B * HP);  // [4*hs*B]
   float* c_local = s_gates + (size_t)4 * hs * B;              // [B*hs]
   __shared__ int lsense;
   if (threadIdx.x == 0) lsense = 0;
@@ -110,17 +113,24 @@ __global__ __launch_bounds__(kNT) void lstm4_fwd_kernel(
   const int64_t BH = (int64_t)B * H;
   const int G4 = 4 * H;
 
-  // Stage this workgroup's W_hh rows (gate-major slices) once.
+  // Stage this workgroup's W_hh rows (gate-major slices) once; zero the
+  // row pad so vectorized dots read zeros past H.
   if (rs > 0) {
-    for (int i = tid; i < 4 * rs * H; i += kNT) {
-      const int row = i / H;          // g*rs + r
+    for (int i = tid; i < 4 * rs * HP; i += kNT) {
+      const int row = i / HP;          // g*rs + r
       const int g = row / rs, r = row % rs;
-      sW[(g * hs + r) * H + i % H] = w_hh[(int64_t)(g * H + r0 + r) * H + i % H];
+      const int h = i % HP;
+      sW[(g * hs + r) * HP + h] =
+          (h < H) ? w_hh[(int64_t)(g * H + r0 + r) * H + h] : (__bf16)0.f;
     }
     for (int i = tid; i < B * rs; i += kNT) {
       const int b = i / rs, r = i % rs;
       c_local[b * hs + r] = c0[(int64_t)b * H + r0 + r];
     }
+  }
+  // Zero the s_hm row pads once (stage only writes h < H).
+  for (int b = 0; b < B; ++b) {
+    for (int h = H + tid; h < HP; h += kNT) s_hm[(size_t)b * HP + h] = (__bf16)0.f;
   }
   __syncthreads();
 
@@ -128,14 +138,28 @@ __global__ __launch_bounds__(kNT) void lstm4_fwd_kernel(
   for (int t = 0; t < T; ++t) {
     const float* nd_t = notdone + (int64_t)t * B;
     // ---- stage masked h into LDS (and the hm stash, written by block 0).
+    // 8-deep manual prefetch: scoped loads bypass L2, so un-pipelined
+    // serial loads would each pay the full coherence-point latency.
     {
       const float* src = hglob + (int64_t)buf * BH;
       float* hm_t = hm + (int64_t)t * BH;
-      for (int i = tid; i < (int)BH; i += kNT) {
-        const int b = i / H;
-        const float v = scoped_load(&src[i]) * nd_t[b];
-        s_hm[i] = (__bf16)v;
-        if (blockIdx.x == 0) hm_t[i] = v;
+      const int total = (int)BH;
+      for (int base = tid * 8; base < total; base += kNT * 8) {
+        float v[8];
+        const int n = min(8, total - base);
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          if (u < n) v[u] = scoped_load(&src[base + u]);
+        }
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          if (u < n) {
+            const int i = base + u;
+            const float m = v[u] * nd_t[i / H];
+            s_hm[(size_t)(i / H) * HP + i % H] = (__bf16)m;
+            if (blockIdx.x == 0) hm_t[i] = m;
+          }
+        }
       }
     }
     __syncthreads();
@@ -149,10 +173,16 @@ __global__ __launch_bounds__(kNT) void lstm4_fwd_kernel(
         const int b = i / jn4;
         const int jl = i % jn4;
         const int g = jl / rs, r = jl % rs;
-        const __bf16* wrow = sW + (size_t)(g * hs + r) * H;
-        const __bf16* hrow = s_hm + (size_t)b * H;
+        const __bf16* wrow = sW + (size_t)(g * hs + r) * HP;
+        const __bf16* hrow = s_hm + (size_t)b * HP;
         float acc = pre_t[(int64_t)b * G4 + g * H + r0 + r];
-        for (int h = 0; h < H; ++h) acc += bf2f(hrow[h]) * bf2f(wrow[h]);
+        // Vectorized LDS reads (b128): rows are 16 B-padded; pads are zero.
+        for (int h8 = 0; h8 < HP; h8 += 8) {
+          const bf16x8v hv = *reinterpret_cast<const bf16x8v*>(&hrow[h8]);
+          const bf16x8v wv = *reinterpret_cast<const bf16x8v*>(&wrow[h8]);
+#pragma unroll
+          for (int u = 0; u < 8; ++u) acc += bf2f(hv[u]) * bf2f(wv[u]);
+        }
         acc = (g == 2) ? tanhf(acc) : 1.f / (1.f + __expf(-acc));
         s_gates[(g * hs + r) * B + b] = acc;
         gates_t[(int64_t)b * G4 + g * H + r0 + r] = acc;
@@ -215,12 +245,13 @@ __global__ __launch_bounds__(kNT) void lstm4_bwd_kernel(
     int* __restrict__ bar,
     float* __restrict__ dgates,  // [T,B,4H] PRE-activation grads (stash)
     float* __restrict__ dh_out, float* __restrict__ dc_out) {  // [B,H]
+  const int G4P = (4 * H + 7) & ~7;  // padded rows for b128 reads
   extern __shared__ char smem[];
-  __bf16* sWT = reinterpret_cast<__bf16*>(smem);          // [hs][4H]
-  __bf16* s_dg = sWT + (size_t)hs * 4 * H;                // [kDB][4H]
+  __bf16* sWT = reinterpret_cast<__bf16*>(smem);          // [hs][G4P]
+  __bf16* s_dg = sWT + (size_t)hs * G4P;                  // [kDB][G4P]
   float* s_part =
-      reinterpret_cast<float*>(s_dg + (size_t)kDB * 4 * H);  // [kDB*hs*kSEG]
-  float* dc_local = s_part + (size_t)kDB * hs * kSEG;        // [B*hs]
+      reinterpret_cast<float*>(s_dg + (size_t)kDB * G4P);  // [kDB*hs*kSEG]
+  float* dc_local = s_part + (size_t)kDB * hs * kSEG;      // [B*hs]
   __shared__ int lsense;
   if (threadIdx.x == 0) lsense = 0;
 
@@ -231,13 +262,21 @@ __global__ __launch_bounds__(kNT) void lstm4_bwd_kernel(
   const int G4 = 4 * H;
 
   if (rs > 0) {
-    for (int i = tid; i < rs * G4; i += kNT) {
-      const int r = i / G4;
-      sWT[(size_t)r * G4 + i % G4] = w_hh_t[(int64_t)(r0 + r) * G4 + i % G4];
+    for (int i = tid; i < rs * G4P; i += kNT) {
+      const int r = i / G4P;
+      const int j = i % G4P;
+      sWT[(size_t)r * G4P + j] =
+          (j < G4) ? w_hh_t[(int64_t)(r0 + r) * G4 + j] : (__bf16)0.f;
     }
     for (int i = tid; i < B * rs; i += kNT) {
       const int b = i / rs, r = i % rs;
       dc_local[b * hs + r] = d_cT[(int64_t)b * H + r0 + r];
+    }
+  }
+  // Zero the s_dg row pads once (chunk staging only writes j < G4).
+  for (int b = 0; b < kDB; ++b) {
+    for (int j = G4 + tid; j < G4P; j += kNT) {
+      s_dg[(size_t)b * G4P + j] = (__bf16)0.f;
     }
   }
   __syncthreads();
@@ -290,13 +329,31 @@ __global__ __launch_bounds__(kNT) void lstm4_bwd_kernel(
     // for this WG's h'-rows, chunked over b with a per-dot K-split.
     {
       float* dst = dhglob + (int64_t)(buf ^ 1) * BH;
-      const int seg_len = (G4 + kSEG - 1) / kSEG;
+      // Segment bounds are 8-aligned so segment dots use b128 reads.
+      const int seg_len = ((G4P / 8 + kSEG - 1) / kSEG) * 8;
       for (int b0 = 0; b0 < B; b0 += kDB) {
         const int db = min(kDB, B - b0);
-        for (int i = tid; i < db * G4; i += kNT) {
-          s_dg[(i / G4) * G4 + i % G4] =
-              (__bf16)scoped_load(&dgates_comm[(int64_t)(b0 + i / G4) * G4 +
-                                               i % G4]);
+        {
+          const int total = db * G4;
+          for (int base = tid * 8; base < total; base += kNT * 8) {
+            float v[8];
+            const int n = min(8, total - base);
+#pragma unroll
+            for (int u = 0; u < 8; ++u) {
+              if (u < n) {
+                const int i = base + u;
+                v[u] = scoped_load(
+                    &dgates_comm[(int64_t)(b0 + i / G4) * G4 + i % G4]);
+              }
+            }
+#pragma unroll
+            for (int u = 0; u < 8; ++u) {
+              if (u < n) {
+                const int i = base + u;
+                s_dg[(size_t)(i / G4) * G4P + i % G4] = (__bf16)v[u];
+              }
+            }
+          }
         }
         __syncthreads();
         if (rs > 0) {
@@ -305,11 +362,16 @@ __global__ __launch_bounds__(kNT) void lstm4_bwd_kernel(
             const int r = (u / kSEG) % rs;
             const int bl = u / (kSEG * rs);
             const int j0 = seg * seg_len;
-            const int j1 = min(G4, j0 + seg_len);
-            const __bf16* dgrow = s_dg + (size_t)bl * G4;
-            const __bf16* wtr = sWT + (size_t)r * G4;
+            const int j1 = min(G4P, j0 + seg_len);
+            const __bf16* dgrow = s_dg + (size_t)bl * G4P;
+            const __bf16* wtr = sWT + (size_t)r * G4P;
             float acc = 0.f;
-            for (int j = j0; j < j1; ++j) acc += bf2f(dgrow[j]) * bf2f(wtr[j]);
+            for (int j8 = j0; j8 < j1; j8 += 8) {
+              const bf16x8v dv = *reinterpret_cast<const bf16x8v*>(&dgrow[j8]);
+              const bf16x8v wv = *reinterpret_cast<const bf16x8v*>(&wtr[j8]);
+#pragma unroll
+              for (int uu = 0; uu < 8; ++uu) acc += bf2f(dv[uu]) * bf2f(wv[uu]);
+            }
             s_part[u] = acc;
           }
         }
@@ -361,9 +423,11 @@ Lstm4Geometry lstm4_geometry(int B, int H) {
   Lstm4Geometry g;
   g.hs = std::max(1, (H + 63) / 64);
   g.nblocks = (H + g.hs - 1) / g.hs;
-  g.fwd_lds = (size_t)4 * g.hs * H * 2 + (size_t)B * H * 2 +
+  const int HP = (H + 7) & ~7;
+  g.fwd_lds = (size_t)4 * g.hs * HP * 2 + (size_t)B * HP * 2 +
               (size_t)4 * g.hs * B * 4 + (size_t)B * g.hs * 4;
-  g.bwd_lds = (size_t)g.hs * 4 * H * 2 + (size_t)kDB * 4 * H * 2 +
+  const int G4P = (4 * H + 7) & ~7;
+  g.bwd_lds = (size_t)g.hs * G4P * 2 + (size_t)kDB * G4P * 2 +
               (size_t)kDB * g.hs * kSEG * 4 + (size_t)B * g.hs * 4;
   return g;
 }
