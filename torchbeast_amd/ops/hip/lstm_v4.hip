@@ -35,6 +35,7 @@
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
 
+#include <cstdlib>
 #include <vector>
 
 namespace tbamd {
@@ -422,6 +423,10 @@ struct Lstm4Geometry {
 Lstm4Geometry lstm4_geometry(int B, int H) {
   Lstm4Geometry g;
   g.hs = std::max(1, (H + 63) / 64);
+  if (const char* e = std::getenv("TBAMD_LSTM_HS")) {
+    const int v = std::atoi(e);
+    if (v > 0) g.hs = v;  // experimental workgroup-count override
+  }
   g.nblocks = (H + g.hs - 1) / g.hs;
   const int HP = (H + 7) & ~7;
   g.fwd_lds = (size_t)4 * g.hs * HP * 2 + (size_t)B * HP * 2 +
