@@ -422,7 +422,10 @@ struct Lstm4Geometry {
 
 Lstm4Geometry lstm4_geometry(int B, int H) {
   Lstm4Geometry g;
-  g.hs = std::max(1, (H + 63) / 64);
+  // Measured sweep on MI355X (profiles/PROFILE_r2.md): ~173 workgroups is
+  // the sweet spot for H=519 (more parallelism beats fewer barriers; the
+  // kernel is latency- not bandwidth-bound). hs = slice rows per WG.
+  g.hs = std::max(1, (H + 172) / 173);
   if (const char* e = std::getenv("TBAMD_LSTM_HS")) {
     const int v = std::atoi(e);
     if (v > 0) g.hs = v;  // experimental workgroup-count override
