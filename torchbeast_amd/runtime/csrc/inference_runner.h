@@ -17,6 +17,8 @@
 #include <torch/extension.h>
 
 #include <atomic>
+#include <chrono>
+#include <cstdlib>
 #include <mutex>
 #include <set>
 #include <thread>
@@ -108,7 +110,16 @@ class InferenceRunner {
   }
 
   void serve(DynamicBatcher::Batch& batch, at::cuda::CUDAStream& stream) {
+    // Stage timers (enabled by TBAMD_SERVE_TIMINGS): cat / fwd / d2h+sync.
+    const bool timing = serve_timing_;
+    auto now_us = []() {
+      return std::chrono::duration_cast<std::chrono::microseconds>(
+                 std::chrono::steady_clock::now().time_since_epoch())
+          .count();
+    };
+    int64_t t0 = timing ? now_us() : 0;
     TensorNest inputs = batch.get_inputs();
+    int64_t t_cat = timing ? now_us() : 0;
     // Serialize the FIRST forward of each quantized batch size: concurrent
     // MIOpen solution-finds for a brand-new conv shape across streams have
     // been observed to fault; once found, the solution cache makes later
@@ -273,12 +284,27 @@ class InferenceRunner {
       out.copy_(t, /*non_blocking=*/true);
       return out;
     };
+    int64_t t_fwd = timing ? now_us() : 0;
     torch::Tensor a_h = to_host(action);
     torch::Tensor l_h = to_host(logits);
     torch::Tensor b_h = to_host(baseline);
     std::vector<torch::Tensor> state_h;
     for (const auto& s : new_state_gpu) state_h.push_back(to_host(s));
     stream.synchronize();
+    if (timing) {
+      const int64_t t_sync = now_us();
+      t_cat_us_.fetch_add(t_cat - t0, std::memory_order_relaxed);
+      t_fwd_us_.fetch_add(t_fwd - t_cat, std::memory_order_relaxed);
+      t_d2h_us_.fetch_add(t_sync - t_fwd, std::memory_order_relaxed);
+      const int64_t n = timed_batches_.fetch_add(1) + 1;
+      if (n % 500 == 0) {
+        fprintf(stderr,
+                "[serve timings over %lld batches] cat=%.0fus fwd=%.0fus "
+                "d2h+sync=%.0fus\n",
+                (long long)n, (double)t_cat_us_.load() / n,
+                (double)t_fwd_us_.load() / n, (double)t_d2h_us_.load() / n);
+      }
+    }
 
     TensorNest::vector_t agent_out{
         TensorNest(a_h.reshape({1, b})),
@@ -298,6 +324,9 @@ class InferenceRunner {
     }
   }
 
+  const bool serve_timing_ = std::getenv("TBAMD_SERVE_TIMINGS") != nullptr;
+  std::atomic<int64_t> t_cat_us_{0}, t_fwd_us_{0}, t_d2h_us_{0};
+  std::atomic<int64_t> timed_batches_{0};
   std::mutex warm_mu_;
   std::mutex serve_mu_;
   std::set<int64_t> warmed_sizes_;
