@@ -64,7 +64,7 @@ def parse_args():
 
 
 def learner_step(flags, batch_tensors, model, optimizer, scheduler, reducer,
-                 actor_flat, flat_param):
+                 actor_flat, flat_param, inference_runner=None):
     env_outputs = pbl.EnvOutput._make(batch_tensors[:5])
     actor_outputs = pbl.AgentOutput._make(batch_tensors[5:8])
     initial_agent_state = batch_tensors[8:]
@@ -110,6 +110,8 @@ def learner_step(flags, batch_tensors, model, optimizer, scheduler, reducer,
     scheduler.step()
     with torch.no_grad():
         actor_flat.copy_(flat_param)
+    if inference_runner is not None:
+        inference_runner.mark_weights_dirty()
     return total_loss
 
 
@@ -172,21 +174,26 @@ def main():
     initial_agent_state = tuple(
         t.cpu() for t in model.initial_state(batch_size=1)
     )
+    use_cpp_inference = use_cuda and not args.py_inference
     pool = runtime.ActorPool(
         unroll_length=T, learner_queue=learner_queue,
         inference_batcher=inference_batcher,
         env_server_addresses=addresses,
         initial_agent_state=initial_agent_state,
         seed_base=rank * args.actors,  # distinct env streams per rank
+        use_obs_slab=use_cpp_inference,
     )
     pool_thread = threading.Thread(target=pool.run, daemon=True)
     pool_thread.start()
 
     inference_runner = None
-    if use_cuda and not args.py_inference:
+    if use_cpp_inference:
         inference_runner = pbl.make_inference_runner(
             actor_model, inference_batcher
         )
+        slab = pool.obs_slab()  # blocks until the first env observation
+        if slab:
+            inference_runner.set_obs_slab(*slab)
         inference_runner.start(args.num_inference_threads)
     else:
         for _ in range(args.num_inference_threads):
@@ -205,7 +212,7 @@ def main():
     # Warmup (fills the pipeline, compiles/caches kernels).
     for _ in range(args.warmup):
         learner_step(flags, get_batch(queue_iter), model, optimizer, scheduler,
-                     reducer, actor_flat, flat_param)
+                     reducer, actor_flat, flat_param, inference_runner)
 
     timings = None
     if os.environ.get("TBAMD_BENCH_TIMINGS"):
@@ -222,14 +229,15 @@ def main():
     for _ in range(args.steps):
         if timings is None:
             learner_step(flags, get_batch(queue_iter), model, optimizer,
-                         scheduler, reducer, actor_flat, flat_param)
+                         scheduler, reducer, actor_flat, flat_param,
+                         inference_runner)
         else:
             timings.reset()
             batch = get_batch(queue_iter)
             torch.cuda.synchronize()
             timings.time("get_batch")
             learner_step(flags, batch, model, optimizer, scheduler, reducer,
-                         actor_flat, flat_param)
+                         actor_flat, flat_param, inference_runner)
             torch.cuda.synchronize()
             timings.time("learn")
 

@@ -32,6 +32,7 @@ ext_modules = [
             "torchbeast_amd/runtime/csrc/module.cc",
             "torchbeast_amd/ops/hip/atari_trunk.hip",
             "torchbeast_amd/ops/hip/conv_mfma.hip",
+            "torchbeast_amd/runtime/csrc/runtime_kernels.hip",
         ],
         extra_compile_args={
             "cxx": ["-O3", "-std=c++17", "-pthread"] + _san_flags,

@@ -259,3 +259,58 @@ def test_tcp_env_roundtrip():
         assert frames.shape[0] == 7  # T+1
     finally:
         server.stop()
+
+
+def test_obs_slab_slot_requests():
+    """Slot-id request mode: actors publish observations into the pinned
+    slab; inference sees only slot ids. (On GPU the C++ engine gathers the
+    slab device-side; here a Python closure reads it directly.)"""
+    L = 6
+    learner_queue = runtime.BatchingQueue(
+        batch_dim=1, minimum_batch_size=1, maximum_batch_size=1
+    )
+    batcher = runtime.DynamicBatcher(batch_dim=1, minimum_batch_size=1,
+                                     maximum_batch_size=64, timeout_ms=5)
+    pool = runtime.ActorPool(
+        unroll_length=4,
+        learner_queue=learner_queue,
+        inference_batcher=batcher,
+        env_server_addresses=["synthetic:1x8x16:3:%d" % L] * 3,
+        initial_agent_state=(),
+        use_obs_slab=True,
+    )
+    pool_thread = threading.Thread(target=pool.run, daemon=True)
+    pool_thread.start()
+    slab_frames, slab_rew, slab_done = pool.obs_slab()
+    assert slab_frames.shape == (3, 1, 8, 16)
+    if torch.cuda.is_available():
+        assert slab_frames.is_pinned()
+    seen_ids = set()
+
+    def inference():
+        try:
+            for batch in batcher:
+                ids, agent_state = batch.get_inputs()
+                assert ids.dtype == torch.int32 and ids.dim() == 2
+                b = ids.shape[1]
+                for i in ids[0].tolist():
+                    seen_ids.add(i)
+                action = torch.zeros((1, b), dtype=torch.int64)
+                logits = torch.ones((1, b, 3))
+                baseline = torch.zeros((1, b))
+                batch.set_outputs(((action, logits, baseline), agent_state))
+        except runtime.ClosedBatchingQueue:
+            pass
+
+    inf_thread = threading.Thread(target=inference, daemon=True)
+    inf_thread.start()
+    it = iter(learner_queue)
+    rollouts = [next(it) for _ in range(3)]
+    batcher.close()
+    learner_queue.close()
+    pool_thread.join(5)
+    inf_thread.join(5)
+    # All three actors used their own slot; rollouts still carry full frames.
+    assert seen_ids == {0, 1, 2}
+    (env_outputs, _agent), _state = rollouts[0]
+    assert env_outputs[0].shape[0] == 5  # [T+1, 1, ...] frames

@@ -207,6 +207,7 @@ def learn(
     reducer,
     num_updates,
     update_counter,
+    inference_runner=None,
     lock=threading.Lock(),  # noqa: B008
 ):
     """Learner loop: exactly num_updates optimizer steps (shared across
@@ -284,6 +285,8 @@ def learn(
             # Behavior-model sync: one flat copy (D2D on GPU).
             with torch.no_grad():
                 actor_flat.copy_(flat_param)
+            if inference_runner is not None:
+                inference_runner.mark_weights_dirty()
 
             episode_returns = env_outputs.episode_return[env_outputs.done]
             stats["step"] = (stats.get("step", 0)
@@ -416,6 +419,10 @@ def train(flags):  # noqa: C901
     initial_agent_state = model.initial_state(batch_size=1)
     initial_agent_state = tuple(t.cpu() for t in initial_agent_state)
 
+    use_cpp_inference = (
+        flags.actor_device.type == "cuda"
+        and not getattr(flags, "py_inference", False)
+    )
     actor_pool = runtime.ActorPool(
         unroll_length=flags.unroll_length,
         learner_queue=learner_queue,
@@ -423,6 +430,9 @@ def train(flags):  # noqa: C901
         env_server_addresses=addresses,
         initial_agent_state=initial_agent_state,
         seed_base=rank * flags.num_actors,
+        # With the C++ engine, actors publish observations into a pinned
+        # slab and requests carry only slot ids (GPU-side gather).
+        use_obs_slab=use_cpp_inference,
     )
 
     def run_pool():
@@ -438,26 +448,22 @@ def train(flags):  # noqa: C901
     done_so_far = stats.get("step", 0) // (flags.unroll_length * flags.batch_size)
     update_counter = {"mutex": threading.Lock(), "done": done_so_far}
 
+    inference_runner = None
+    if use_cpp_inference:
+        inference_runner = make_inference_runner(actor_model, inference_batcher)
+        inference_threads = []
     learner_threads = [
         threading.Thread(
             target=learn,
             name=f"learner-{i}",
             args=(flags, learner_queue, model, flat_param, flat_grad, actor_flat,
                   optimizer, scheduler, stats, plogger, reducer, num_updates,
-                  update_counter),
+                  update_counter, inference_runner),
         )
         # Collectives must stay ordered: one learner thread under DP.
         for i in range(1 if world_size > 1 else flags.num_learner_threads)
     ]
-    use_cpp_inference = (
-        flags.actor_device.type == "cuda"
-        and not getattr(flags, "py_inference", False)
-    )
-    inference_runner = None
-    if use_cpp_inference:
-        inference_runner = make_inference_runner(actor_model, inference_batcher)
-        inference_threads = []
-    else:
+    if not use_cpp_inference:
         inference_threads = [
             threading.Thread(
                 target=inference,
@@ -469,6 +475,9 @@ def train(flags):  # noqa: C901
 
     actorpool_thread.start()
     if inference_runner is not None:
+        slab = actor_pool.obs_slab()  # blocks until the first env obs
+        if slab:
+            inference_runner.set_obs_slab(*slab)
         inference_runner.start(flags.num_inference_threads)
     for t in learner_threads + inference_threads:
         t.start()

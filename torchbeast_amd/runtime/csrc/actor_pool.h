@@ -10,7 +10,10 @@
 #pragma once
 
 #include <atomic>
+#include <condition_variable>
+#include <cstring>
 #include <future>
+#include <mutex>
 #include <memory>
 #include <string>
 #include <vector>
@@ -25,16 +28,33 @@ class ActorPool {
   ActorPool(int64_t unroll_length, std::shared_ptr<BatchingQueue> learner_queue,
             std::shared_ptr<DynamicBatcher> inference_batcher,
             std::vector<std::string> env_server_addresses,
-            TensorNest initial_agent_state, int64_t seed_base = 0)
+            TensorNest initial_agent_state, int64_t seed_base = 0,
+            bool use_obs_slab = false)
       : unroll_length_(unroll_length),
         learner_queue_(std::move(learner_queue)),
         inference_batcher_(std::move(inference_batcher)),
         addresses_(std::move(env_server_addresses)),
         initial_agent_state_(std::move(initial_agent_state)),
-        seed_base_(seed_base) {
+        seed_base_(seed_base),
+        use_obs_slab_(use_obs_slab) {
     if (unroll_length_ < 1) {
       throw std::invalid_argument("unroll_length must be >= 1");
     }
+  }
+
+  // Pinned observation slab (slot per actor). Valid once the first env has
+  // produced an observation; the inference runner gathers from it by slot
+  // id instead of cat-ing per-request frame tensors.
+  std::vector<torch::Tensor> obs_slab() {
+    std::unique_lock<std::mutex> lk(slab_mu_);
+    slab_cv_.wait(lk, [this] {
+      return slab_ready_ || failed_ || !use_obs_slab_;
+    });
+    if (failed_ && !slab_ready_) {
+      throw std::runtime_error("actor pool failed before producing obs");
+    }
+    if (!use_obs_slab_) return {};
+    return {slab_frames_, slab_rew_, slab_done_};
   }
 
   // Blocks until every actor thread exits (via queue close or error).
@@ -43,8 +63,19 @@ class ActorPool {
     std::vector<std::future<void>> futures;
     futures.reserve(addresses_.size());
     for (size_t i = 0; i < addresses_.size(); ++i) {
-      futures.push_back(std::async(
-          std::launch::async, [this, i] { loop(addresses_[i], i); }));
+      futures.push_back(std::async(std::launch::async, [this, i] {
+        try {
+          loop(addresses_[i], i);
+        } catch (...) {
+          // Wake obs_slab() waiters so setup can't hang on a dead actor.
+          {
+            std::lock_guard<std::mutex> lk(slab_mu_);
+            failed_ = true;
+          }
+          slab_cv_.notify_all();
+          throw;
+        }
+      }));
     }
     std::exception_ptr first_error;
     for (auto& f : futures) {
@@ -62,6 +93,37 @@ class ActorPool {
   uint64_t count() const { return step_count_.load(std::memory_order_relaxed); }
 
  private:
+  // Copy this actor's observation into its pinned slot; the request nest
+  // then carries only the slot id (gathered GPU-side by the runner).
+  void fill_slot(int64_t slot, const TensorNest& env_outputs) {
+    const auto& f = env_outputs.vector();
+    const torch::Tensor frame = f[0].leaf();
+    if (!slab_ready_) {
+      std::lock_guard<std::mutex> lk(slab_mu_);
+      if (!slab_ready_) {
+        const int64_t n = (int64_t)addresses_.size();
+        auto fshape = frame.sizes().vec();  // [1,1,C,H,W]
+        std::vector<int64_t> slab_shape = {n};
+        slab_shape.insert(slab_shape.end(), fshape.begin() + 2, fshape.end());
+        // Pinned only when a GPU exists (pinned allocs require HIP);
+        // CPU runs use the slab purely as shared host storage.
+        auto hopts =
+            torch::TensorOptions().pinned_memory(torch::cuda::is_available());
+        slab_frames_ = torch::empty(slab_shape, hopts.dtype(torch::kUInt8));
+        slab_rew_ = torch::zeros({n}, hopts.dtype(torch::kFloat32));
+        slab_done_ = torch::zeros({n}, hopts.dtype(torch::kUInt8));
+        slab_ready_ = true;
+        slab_cv_.notify_all();
+      }
+    }
+    const int64_t fsz = slab_frames_.stride(0);
+    std::memcpy(
+        static_cast<uint8_t*>(slab_frames_.data_ptr()) + slot * fsz,
+        frame.data_ptr(), fsz);
+    slab_rew_.data_ptr<float>()[slot] = f[1].leaf().item<float>();
+    slab_done_.data_ptr<uint8_t>()[slot] = f[2].leaf().item<bool>() ? 1 : 0;
+  }
+
   void loop(const std::string& address, uint64_t seed) {
     auto env = make_env_connection(address, seed_base_ + seed + 1);
 
@@ -69,13 +131,27 @@ class ActorPool {
     TensorNest agent_state = initial_agent_state_;
     TensorNest rollout_initial_state = initial_agent_state_;
 
+    // Slot-id request leaf (constant per actor, reused every step).
+    TensorNest slot_req;
+    if (use_obs_slab_) {
+      slot_req = TensorNest(torch::full({1, 1}, (int64_t)seed,
+                                        torch::TensorOptions().dtype(
+                                            torch::kInt32)));
+    }
+
     std::vector<TensorNest> rollout;
     rollout.reserve(unroll_length_ + 1);
 
     for (;;) {
       TensorNest state_before = agent_state;
-      TensorNest result = inference_batcher_->compute(
-          TensorNest(TensorNest::vector_t{env_outputs, agent_state}));
+      TensorNest request;
+      if (use_obs_slab_) {
+        fill_slot((int64_t)seed, env_outputs);
+        request = TensorNest(TensorNest::vector_t{slot_req, agent_state});
+      } else {
+        request = TensorNest(TensorNest::vector_t{env_outputs, agent_state});
+      }
+      TensorNest result = inference_batcher_->compute(std::move(request));
       if (!result.is_vector() || result.vector().size() != 2) {
         throw std::runtime_error(
             "inference must return ((action, ...), agent_state)");
@@ -119,6 +195,12 @@ class ActorPool {
   std::vector<std::string> addresses_;
   TensorNest initial_agent_state_;
   const int64_t seed_base_;
+  const bool use_obs_slab_;
+  std::mutex slab_mu_;
+  std::condition_variable slab_cv_;
+  std::atomic<bool> slab_ready_{false};
+  bool failed_ = false;
+  torch::Tensor slab_frames_, slab_rew_, slab_done_;
   std::atomic<uint64_t> step_count_{0};
 };
 

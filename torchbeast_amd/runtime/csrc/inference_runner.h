@@ -27,6 +27,7 @@
 #include "../../ops/hip/atari_trunk.h"
 #include "../../ops/hip/conv_mfma.h"
 #include "queues.h"
+#include "runtime_kernels.h"
 
 namespace tbruntime {
 
@@ -83,6 +84,23 @@ class InferenceRunner {
   int64_t batches() const { return batches_.load(); }
   int64_t steps() const { return steps_.load(); }
 
+  // Observation slab (from ActorPool::obs_slab) enabling the slot-id fast
+  // path: requests carry only actor ids; frames/reward/done are gathered
+  // GPU-side straight from the pinned slab.
+  void set_obs_slab(torch::Tensor frames, torch::Tensor rew,
+                    torch::Tensor done) {
+    slab_frames_ = std::move(frames);
+    slab_rew_ = std::move(rew);
+    slab_done_ = std::move(done);
+    frame_shape_ = std::vector<int64_t>(slab_frames_.sizes().begin() + 1,
+                                        slab_frames_.sizes().end());
+    has_slab_ = true;
+  }
+
+  // Learner weight sync happened: drop the cached bf16-packed trunk
+  // weights (repacked lazily by the next serve).
+  void mark_weights_dirty() { weights_version_.fetch_add(1); }
+
  private:
   void loop() {
     c10::cuda::CUDAGuard device_guard(device_);
@@ -134,34 +152,52 @@ class InferenceRunner {
     if (is_new) {
       warm_lock = std::unique_lock<std::mutex>(serve_mu_);
     }
-    // inputs = ((frame, reward, done, episode_step, episode_return), state)
+    // inputs = ((frame, reward, done, ...), state)  [classic requests]
+    //        = (slot_ids, state)                     [obs-slab requests]
     const auto& top = inputs.vector();
-    const auto& env = top[0].vector();
-    const torch::Tensor frame = env[0].leaf();    // [1, b, C, H, W] u8
-    const torch::Tensor reward = env[1].leaf();   // [1, b] f32
-    const torch::Tensor done = env[2].leaf();     // [1, b] bool
+    const bool slot_mode = top[0].is_leaf();
     std::vector<torch::Tensor> state = top[1].flatten();  // [] or h,c [L,b,H]
-
-    const int64_t b = frame.size(1);
-    // Quantize the compute batch to multiples of 64: MIOpen caches conv
-    // solutions per shape, and ragged dynamic-batch sizes would trigger a
-    // fresh find for every new size. Pad rows are zeroed and sliced off
-    // before sampling/D2H.
-    const int64_t bp = (b + 63) / 64 * 64;
     auto opts = torch::TensorOptions().device(device_);
 
-    const int64_t C = frame.size(2), H = frame.size(3), W = frame.size(4);
-    torch::Tensor frames_p =
-        torch::empty({bp, C, H, W}, opts.dtype(torch::kUInt8));
-    frames_p.narrow(0, 0, b).copy_(frame.reshape({b, C, H, W}),
-                                   /*non_blocking=*/true);
-    if (bp > b) frames_p.narrow(0, b, bp - b).zero_();
-
-    torch::Tensor rew = torch::empty({bp, 1}, opts.dtype(torch::kFloat32));
-    rew.narrow(0, 0, b)
-        .copy_(reward.to(opts.dtype(torch::kFloat32), true).reshape({b, 1}))
-        .clamp_(-1, 1);
-    if (bp > b) rew.narrow(0, b, bp - b).zero_();
+    int64_t b, bp, C, H, W;
+    torch::Tensor frames_p, rew, nd_gpu;
+    if (slot_mode) {
+      TORCH_CHECK(has_slab_, "slot-id request but no obs slab configured");
+      const torch::Tensor ids = top[0].leaf();  // [1, b] int32
+      b = ids.size(1);
+      bp = (b + 63) / 64 * 64;
+      auto g = gather_obs(slab_frames_, slab_rew_, slab_done_, ids, bp,
+                          frame_shape_);
+      frames_p = g[0];
+      rew = g[1];
+      nd_gpu = g[2];
+      C = frame_shape_[0];
+      H = frame_shape_[1];
+      W = frame_shape_[2];
+    } else {
+      const auto& env = top[0].vector();
+      const torch::Tensor frame = env[0].leaf();   // [1, b, C, H, W] u8
+      const torch::Tensor reward = env[1].leaf();  // [1, b] f32
+      const torch::Tensor done = env[2].leaf();    // [1, b] bool
+      b = frame.size(1);
+      // Quantize the compute batch to multiples of 64 so kernel/solution
+      // caches see a handful of shapes, not every ragged batch size.
+      bp = (b + 63) / 64 * 64;
+      C = frame.size(2);
+      H = frame.size(3);
+      W = frame.size(4);
+      frames_p = torch::empty({bp, C, H, W}, opts.dtype(torch::kUInt8));
+      frames_p.narrow(0, 0, b).copy_(frame.reshape({b, C, H, W}),
+                                     /*non_blocking=*/true);
+      if (bp > b) frames_p.narrow(0, b, bp - b).zero_();
+      rew = torch::empty({bp, 1}, opts.dtype(torch::kFloat32));
+      rew.narrow(0, 0, b)
+          .copy_(reward.to(opts.dtype(torch::kFloat32), true).reshape({b, 1}))
+          .clamp_(-1, 1);
+      if (bp > b) rew.narrow(0, b, bp - b).zero_();
+      nd_gpu = (~done.reshape({b}))
+                   .to(opts.dtype(torch::kFloat32), /*non_blocking=*/true);
+    }
 
     torch::Tensor x;
     if (deep_) {
@@ -186,22 +222,29 @@ class InferenceRunner {
       // fused kernel whose grid is the batch (underfills the 256 CUs below
       // ~512 samples and was measured dominating GPU time at small
       // batches: profiles/PROFILE_r2.md).
-      x = tbamd::conv_trunk_fwd(
-          frames_p,
-          weights_[0].reshape({32, -1}).to(torch::kBFloat16).contiguous(),
-          weights_[1],
-          weights_[2]
-              .permute({0, 2, 3, 1})
-              .reshape({64, -1})
-              .to(torch::kBFloat16)
-              .contiguous(),
-          weights_[3],
-          weights_[4]
-              .permute({0, 2, 3, 1})
-              .reshape({64, -1})
-              .to(torch::kBFloat16)
-              .contiguous(),
-          weights_[5], /*want_stash=*/false)[0];
+      torch::Tensor w1p, w2p, w3p;
+      {
+        // bf16-packed trunk weights, cached until the learner's next sync
+        // (mark_weights_dirty). The packing thread synchronizes its stream
+        // before publishing so other serve streams read complete data.
+        std::lock_guard<std::mutex> g(pack_mu_);
+        const int64_t v = weights_version_.load();
+        if (packed_version_ != v) {
+          w1p_ = weights_[0].reshape({32, -1}).to(torch::kBFloat16)
+                     .contiguous();
+          w2p_ = weights_[2].permute({0, 2, 3, 1}).reshape({64, -1})
+                     .to(torch::kBFloat16).contiguous();
+          w3p_ = weights_[4].permute({0, 2, 3, 1}).reshape({64, -1})
+                     .to(torch::kBFloat16).contiguous();
+          stream.synchronize();
+          packed_version_ = v;
+        }
+        w1p = w1p_;
+        w2p = w2p_;
+        w3p = w3p_;
+      }
+      x = tbamd::conv_trunk_fwd(frames_p, w1p, weights_[1], w2p, weights_[3],
+                                w3p, weights_[5], /*want_stash=*/false)[0];
     } else if (bp <= 384 && tbamd::atari_trunk_supported(C, H, W)) {
       // Hand-written fused CDNA4 conv trunk: one kernel for the u8
       // normalize + 3 convs (non-84x84 geometries).
@@ -217,6 +260,47 @@ class InferenceRunner {
     }
     x = at::linear(x, weights_[head_base_ - 2], weights_[head_base_ - 1])
             .relu_();
+
+    if (num_lstm_layers_ == 0 && weights_[head_base_].size(0) <= 64 &&
+        weights_[head_base_].size(1) == x.size(1) + 1) {
+      // Fused heads + Gumbel sample, written straight into pinned host
+      // buffers: replaces cat/2x linear/rand/log/argmax/3x D2H.
+      const int64_t seed =
+          greedy_ ? 0 : (int64_t)(seed_ctr_.fetch_add(1) * 0x9E3779B97F4A7C15ull);
+      auto hs = fused_heads_sample(
+          x, rew, weights_[head_base_], weights_[head_base_ + 1],
+          weights_[head_base_ + 2], weights_[head_base_ + 3], b, greedy_,
+          seed);
+      stream.synchronize();
+      int64_t t_fwd2 = timing ? now_us() : 0;
+      TensorNest::vector_t agent_out{
+          TensorNest(hs[0].reshape({1, b})),
+          TensorNest(hs[1].reshape({1, b, -1})),
+          TensorNest(hs[2].reshape({1, b})),
+      };
+      batch.set_outputs(TensorNest(TensorNest::vector_t{
+          TensorNest(std::move(agent_out)), TensorNest(TensorNest::vector_t{})}));
+      batches_.fetch_add(1, std::memory_order_relaxed);
+      steps_.fetch_add(b, std::memory_order_relaxed);
+      if (timing) {
+        t_cat_us_.fetch_add(t_cat - t0, std::memory_order_relaxed);
+        t_fwd_us_.fetch_add(t_fwd2 - t_cat, std::memory_order_relaxed);
+        const int64_t n = timed_batches_.fetch_add(1) + 1;
+        if (n % 500 == 0) {
+          fprintf(stderr,
+                  "[serve timings over %lld batches] cat=%.0fus "
+                  "fwd+sync=%.0fus\n",
+                  (long long)n, (double)t_cat_us_.load() / n,
+                  (double)t_fwd_us_.load() / n);
+        }
+      }
+      if (warm_lock.owns_lock()) {
+        std::lock_guard<std::mutex> g(warm_mu_);
+        warmed_sizes_.insert(bq);
+      }
+      return;
+    }
+
     torch::Tensor core = at::cat({x, rew}, 1);
 
     std::vector<torch::Tensor> new_state_gpu;
@@ -224,8 +308,7 @@ class InferenceRunner {
       TORCH_CHECK(state.size() == 2, "lstm runner needs (h, c) state");
       const int64_t L = state[0].size(0);
       const int64_t H = state[0].size(2);
-      torch::Tensor nd = (~done.reshape({1, b, 1}))
-                             .to(opts.dtype(torch::kFloat32), true);
+      torch::Tensor nd = nd_gpu.narrow(0, 0, b).reshape({1, b, 1});
       auto pad_state = [&](const torch::Tensor& s) {
         torch::Tensor g = torch::zeros({L, bp, H}, opts.dtype(torch::kFloat32));
         g.narrow(1, 0, b).copy_(
@@ -324,6 +407,14 @@ class InferenceRunner {
     }
   }
 
+  bool has_slab_ = false;
+  torch::Tensor slab_frames_, slab_rew_, slab_done_;
+  std::vector<int64_t> frame_shape_;
+  std::mutex pack_mu_;
+  torch::Tensor w1p_, w2p_, w3p_;
+  int64_t packed_version_ = -1;
+  std::atomic<int64_t> weights_version_{0};
+  std::atomic<uint64_t> seed_ctr_{1};
   const bool serve_timing_ = std::getenv("TBAMD_SERVE_TIMINGS") != nullptr;
   std::atomic<int64_t> t_cat_us_{0}, t_fwd_us_{0}, t_d2h_us_{0};
   std::atomic<int64_t> timed_batches_{0};

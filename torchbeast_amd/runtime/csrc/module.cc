@@ -237,11 +237,14 @@ PYBIND11_MODULE(_tbruntime, m) {
   py::class_<ActorPool, std::shared_ptr<ActorPool>>(m, "ActorPool")
       .def(py::init<int64_t, std::shared_ptr<BatchingQueue>,
                     std::shared_ptr<DynamicBatcher>, std::vector<std::string>,
-                    TensorNest, int64_t>(),
+                    TensorNest, int64_t, bool>(),
            py::arg("unroll_length"), py::arg("learner_queue"),
            py::arg("inference_batcher"), py::arg("env_server_addresses"),
-           py::arg("initial_agent_state"), py::arg("seed_base") = 0)
+           py::arg("initial_agent_state"), py::arg("seed_base") = 0,
+           py::arg("use_obs_slab") = false)
       .def("run", &ActorPool::run, py::call_guard<py::gil_scoped_release>())
+      .def("obs_slab", &ActorPool::obs_slab,
+           py::call_guard<py::gil_scoped_release>())
       .def("count", &ActorPool::count);
 
   // ---- InferenceRunner ----
@@ -257,7 +260,9 @@ PYBIND11_MODULE(_tbruntime, m) {
       .def("stop", &InferenceRunner::stop,
            py::call_guard<py::gil_scoped_release>())
       .def("batches", &InferenceRunner::batches)
-      .def("steps", &InferenceRunner::steps);
+      .def("steps", &InferenceRunner::steps)
+      .def("set_obs_slab", &InferenceRunner::set_obs_slab)
+      .def("mark_weights_dirty", &InferenceRunner::mark_weights_dirty);
 
   // ---- EnvServer ----
   py::class_<EnvServer, std::shared_ptr<EnvServer>>(m, "Server")
