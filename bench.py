@@ -175,13 +175,14 @@ def main():
         t.cpu() for t in model.initial_state(batch_size=1)
     )
     use_cpp_inference = use_cuda and not args.py_inference
+    use_obs_slab = use_cpp_inference and not os.environ.get("TBAMD_NO_OBS_SLAB")
     pool = runtime.ActorPool(
         unroll_length=T, learner_queue=learner_queue,
         inference_batcher=inference_batcher,
         env_server_addresses=addresses,
         initial_agent_state=initial_agent_state,
         seed_base=rank * args.actors,  # distinct env streams per rank
-        use_obs_slab=use_cpp_inference,
+        use_obs_slab=use_obs_slab,
     )
     pool_thread = threading.Thread(target=pool.run, daemon=True)
     pool_thread.start()
@@ -191,9 +192,10 @@ def main():
         inference_runner = pbl.make_inference_runner(
             actor_model, inference_batcher
         )
-        slab = pool.obs_slab()  # blocks until the first env observation
-        if slab:
-            inference_runner.set_obs_slab(*slab)
+        if use_obs_slab:
+            slab = pool.obs_slab()  # blocks until the first env observation
+            if slab:
+                inference_runner.set_obs_slab(*slab)
         inference_runner.start(args.num_inference_threads)
     else:
         for _ in range(args.num_inference_threads):
