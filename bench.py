@@ -175,7 +175,10 @@ def main():
         t.cpu() for t in model.initial_state(batch_size=1)
     )
     use_cpp_inference = use_cuda and not args.py_inference
-    use_obs_slab = use_cpp_inference and not os.environ.get("TBAMD_NO_OBS_SLAB")
+    # Measured on MI355X: the GPU-side slab gather loses ~12% to the
+    # pinned-cat + SDMA copy path (kernel reads over the host link are
+    # slower than the DMA engine). Opt-in for experiments.
+    use_obs_slab = use_cpp_inference and bool(os.environ.get("TBAMD_OBS_SLAB"))
     pool = runtime.ActorPool(
         unroll_length=T, learner_queue=learner_queue,
         inference_batcher=inference_batcher,

@@ -430,9 +430,10 @@ def train(flags):  # noqa: C901
         env_server_addresses=addresses,
         initial_agent_state=initial_agent_state,
         seed_base=rank * flags.num_actors,
-        # With the C++ engine, actors publish observations into a pinned
-        # slab and requests carry only slot ids (GPU-side gather).
-        use_obs_slab=use_cpp_inference,
+        # Opt-in (TBAMD_OBS_SLAB): actors publish observations into a
+        # pinned slab and requests carry only slot ids (GPU-side gather).
+        # Default off: measured slower than pinned-cat + SDMA on MI355X.
+        use_obs_slab=use_cpp_inference and bool(os.environ.get("TBAMD_OBS_SLAB")),
     )
 
     def run_pool():
@@ -475,9 +476,10 @@ def train(flags):  # noqa: C901
 
     actorpool_thread.start()
     if inference_runner is not None:
-        slab = actor_pool.obs_slab()  # blocks until the first env obs
-        if slab:
-            inference_runner.set_obs_slab(*slab)
+        if os.environ.get("TBAMD_OBS_SLAB"):
+            slab = actor_pool.obs_slab()  # blocks until the first env obs
+            if slab:
+                inference_runner.set_obs_slab(*slab)
         inference_runner.start(flags.num_inference_threads)
     for t in learner_threads + inference_threads:
         t.start()

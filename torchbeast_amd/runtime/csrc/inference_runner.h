@@ -143,14 +143,20 @@ class InferenceRunner {
     // been observed to fault; once found, the solution cache makes later
     // serves safe to run fully in parallel.
     const int64_t bq = (batch.size() + 63) / 64 * 64;
-    bool is_new;
-    {
-      std::lock_guard<std::mutex> g(warm_mu_);
-      is_new = warmed_sizes_.count(bq) == 0;
-    }
     std::unique_lock<std::mutex> warm_lock;
-    if (is_new) {
-      warm_lock = std::unique_lock<std::mutex>(serve_mu_);
+    if (deep_) {
+      // Only the MIOpen (deep ResNet) path needs first-serve-per-shape
+      // serialization: concurrent solution-finds for a new conv shape
+      // across streams have been observed to fault. The hand-written
+      // trunk kernels have no such state.
+      bool is_new;
+      {
+        std::lock_guard<std::mutex> g(warm_mu_);
+        is_new = warmed_sizes_.count(bq) == 0;
+      }
+      if (is_new) {
+        warm_lock = std::unique_lock<std::mutex>(serve_mu_);
+      }
     }
     // inputs = ((frame, reward, done, ...), state)  [classic requests]
     //        = (slot_ids, state)                     [obs-slab requests]
