@@ -58,6 +58,7 @@ def parse_args():
     p.add_argument("--inference_max_batch_size", type=int, default=512)
     p.add_argument("--inference_timeout_ms", type=int, default=5)
     p.add_argument("--episode_length", type=int, default=1000)
+    p.add_argument("--rollout_budget_mb", type=int, default=1024)
     p.add_argument("--py_inference", action="store_true",
                    help="Python inference threads instead of the C++ engine.")
     return p.parse_args()
@@ -186,6 +187,7 @@ def main():
         initial_agent_state=initial_agent_state,
         seed_base=rank * args.actors,  # distinct env streams per rank
         use_obs_slab=use_obs_slab,
+        rollout_budget_mb=args.rollout_budget_mb,
     )
     pool_thread = threading.Thread(target=pool.run, daemon=True)
     pool_thread.start()

@@ -58,6 +58,10 @@ def make_parser():
     parser.add_argument("--num_learner_threads", default=2, type=int)
     parser.add_argument("--num_inference_threads", default=2, type=int)
     parser.add_argument("--max_learner_queue_size", default=None, type=int)
+    parser.add_argument("--rollout_buffer_budget_mb", default=1024, type=int,
+                        help="Pinned rollout ring budget (MB); actors block "
+                             "when it is exhausted (0 = unbounded ad-hoc "
+                             "pinned allocations).")
     parser.add_argument("--num_actions", default=6, type=int)
     parser.add_argument("--model", default="shallow", choices=["shallow", "deep"],
                         help="shallow=AtariNet (the headline bench model), "
@@ -434,6 +438,7 @@ def train(flags):  # noqa: C901
         # pinned slab and requests carry only slot ids (GPU-side gather).
         # Default off: measured slower than pinned-cat + SDMA on MI355X.
         use_obs_slab=use_cpp_inference and bool(os.environ.get("TBAMD_OBS_SLAB")),
+        rollout_budget_mb=flags.rollout_buffer_budget_mb,
     )
 
     def run_pool():

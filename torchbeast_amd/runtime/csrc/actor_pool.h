@@ -29,7 +29,7 @@ class ActorPool {
             std::shared_ptr<DynamicBatcher> inference_batcher,
             std::vector<std::string> env_server_addresses,
             TensorNest initial_agent_state, int64_t seed_base = 0,
-            bool use_obs_slab = false)
+            bool use_obs_slab = false, int64_t rollout_budget_mb = 0)
       : unroll_length_(unroll_length),
         learner_queue_(std::move(learner_queue)),
         inference_batcher_(std::move(inference_batcher)),
@@ -39,6 +39,14 @@ class ActorPool {
         use_obs_slab_(use_obs_slab) {
     if (unroll_length_ < 1) {
       throw std::invalid_argument("unroll_length must be >= 1");
+    }
+    if (rollout_budget_mb > 0) {
+      // Budget-sized pinned rollout ring with backpressure (replaces
+      // unbounded ad-hoc pinned allocations); the learner queue recycles
+      // slots after its H2D assembly completes.
+      rollout_pool_ =
+          std::make_shared<PinnedSlabPool>(rollout_budget_mb * (1 << 20));
+      learner_queue_->set_source_pool(rollout_pool_);
     }
   }
 
@@ -167,11 +175,44 @@ class ActorPool {
         std::vector<const TensorNest*> steps;
         steps.reserve(rollout.size());
         for (const auto& s : rollout) steps.push_back(&s);
-        // Pinned slab so the learner-side dequeue can DMA it to the GPU.
-        TensorNest stacked = TensorNest::apply_columns(
-            steps, [](const std::vector<torch::Tensor>& column) {
-              return cat_pinned(column, /*dim=*/0);
-            });
+        // Pinned destination so the learner-side dequeue can DMA it to the
+        // GPU: a slot of the budget-bounded ring when configured, else an
+        // ad-hoc pinned allocation.
+        TensorNest stacked;
+        if (rollout_pool_) {
+          if (rollout_slot_bytes_ == 0) {
+            int64_t total = 0;
+            TensorNest::apply_columns(
+                steps, [&total](const std::vector<torch::Tensor>& column) {
+                  int64_t rows = 0;
+                  for (const auto& t : column) rows += t.size(0);
+                  const int64_t bytes = rows * column[0].stride(0) *
+                                        column[0].element_size();
+                  total += (bytes + 255) & ~int64_t(255);
+                  return column[0];
+                });
+            rollout_slot_bytes_ = total + 4096;
+          }
+          auto slot = rollout_pool_->acquire(
+              rollout_slot_bytes_,
+              [this] { return learner_queue_->is_closed(); });
+          stacked = TensorNest::apply_columns(
+              steps, [&slot](const std::vector<torch::Tensor>& column) {
+                auto shape = column[0].sizes().vec();
+                int64_t rows = 0;
+                for (const auto& t : column) rows += t.size(0);
+                shape[0] = rows;
+                torch::Tensor out =
+                    slot.carve(shape, column[0].scalar_type());
+                torch::cat_out(out, column, 0);
+                return out;
+              });
+        } else {
+          stacked = TensorNest::apply_columns(
+              steps, [](const std::vector<torch::Tensor>& column) {
+                return cat_pinned(column, /*dim=*/0);
+              });
+        }
         learner_queue_->enqueue(TensorNest(
             TensorNest::vector_t{std::move(stacked), rollout_initial_state}));
 
@@ -201,6 +242,8 @@ class ActorPool {
   std::atomic<bool> slab_ready_{false};
   bool failed_ = false;
   torch::Tensor slab_frames_, slab_rew_, slab_done_;
+  std::shared_ptr<PinnedSlabPool> rollout_pool_;
+  std::atomic<int64_t> rollout_slot_bytes_{0};
   std::atomic<uint64_t> step_count_{0};
 };
 

@@ -18,6 +18,7 @@
 #include <chrono>
 #include <condition_variable>
 #include <deque>
+#include <functional>
 #include <future>
 #include <map>
 #include <memory>
@@ -83,6 +84,157 @@ inline torch::Tensor cat_pinned(const std::vector<torch::Tensor>& tensors,
   torch::cat_out(out, tensors, dim);
   return out;
 }
+
+// ---------------------------------------------------------------------------
+// PinnedSlabPool: fixed-capacity pinned rollout buffering with backpressure.
+//
+// One pinned slab carved into equal rollout-sized slots (slot size fixed by
+// the first acquire; every rollout of a run has the same shape). An actor
+// acquires a slot, cats its finished rollout into slab VIEWS (so
+// .is_pinned() stays true and H2D copies stay DMA), and enqueues them.
+// Rollout tensors never escape the learner-side BatchingQueue (both
+// assembly modes produce fresh batched outputs), so slot recycling is
+// driven by the queue alone: after assembling a batch it calls
+// mark_consumed(ptr, event) per source leaf, and the slot returns to the
+// free list once that dequeue's copy event has completed (immediately in
+// CPU mode). acquire() blocks when the budget is exhausted — this is the
+// backpressure that bounds rollout memory to the configured budget
+// (BASELINE.json north star: buffering sized against a memory budget, not
+// unbounded ad-hoc allocations).
+// ---------------------------------------------------------------------------
+
+class PinnedSlabPool {
+ public:
+  explicit PinnedSlabPool(int64_t budget_bytes, int64_t min_slots = 64)
+      : budget_bytes_(budget_bytes), min_slots_(min_slots) {}
+
+  // Handle for filling one slot; carve() returns pinned slab views.
+  struct Slot {
+    PinnedSlabPool* pool;
+    int64_t index;
+    int64_t used = 0;
+
+    torch::Tensor carve(std::vector<int64_t> shape, torch::ScalarType dtype) {
+      int64_t numel = 1;
+      for (auto d : shape) numel *= d;
+      const int64_t nbytes = numel * (int64_t)torch::elementSize(dtype);
+      const int64_t aligned = (used + 255) & ~int64_t(255);
+      TORCH_CHECK(aligned + nbytes <= pool->slot_bytes_,
+                  "rollout slot overflow");
+      torch::Tensor flat = pool->slab_.narrow(
+          0, index * pool->slot_bytes_ + aligned, nbytes);
+      used = aligned + nbytes;
+      return flat.view(dtype).reshape(shape);
+    }
+  };
+
+  // Blocks until a slot is free; `cancelled` (checked every 50 ms) lets
+  // shutdown unwind the actor threads.
+  Slot acquire(int64_t slot_bytes, const std::function<bool()>& cancelled) {
+    std::unique_lock<std::mutex> lk(mu_);
+    if (!slab_.defined()) {
+      slot_bytes_ = (slot_bytes + 4095) & ~int64_t(4095);
+      int64_t slots = budget_bytes_ / slot_bytes_;
+      if (slots < min_slots_) slots = min_slots_;
+      slab_ = torch::empty(
+          {slots * slot_bytes_},
+          torch::TensorOptions()
+              .dtype(torch::kUInt8)
+              .pinned_memory(pinned_memory_wanted()));
+      for (int64_t i = slots - 1; i >= 0; --i) free_.push_back(i);
+      total_slots_ = slots;
+    }
+    TORCH_CHECK(slot_bytes <= slot_bytes_,
+                "rollout larger than the slab slot size");
+    for (;;) {
+      reap_pending();
+      if (!free_.empty()) {
+        const int64_t idx = free_.back();
+        free_.pop_back();
+        return Slot{this, idx, 0};
+      }
+      backpressure_waits_.fetch_add(1, std::memory_order_relaxed);
+      cv_.wait_for(lk, std::chrono::milliseconds(50));
+      if (cancelled && cancelled()) {
+        throw ClosedQueue("slab pool acquire cancelled");
+      }
+    }
+  }
+
+  // Consumer-side: `ptr` was read by copies whose completion `ev` tracks
+  // (ev == nullptr for synchronous CPU assembly). Idempotent per slot per
+  // dequeue: all leaves of a rollout carry the same event.
+  void mark_consumed(const void* ptr,
+                     const std::shared_ptr<at::cuda::CUDAEvent>& ev) {
+    std::lock_guard<std::mutex> lk(mu_);
+    if (!slab_.defined()) return;
+    const uint8_t* base = slab_.data_ptr<uint8_t>();
+    const auto* p = static_cast<const uint8_t*>(ptr);
+    if (p < base || p >= base + total_slots_ * slot_bytes_) return;
+    const int64_t idx = (p - base) / slot_bytes_;
+    if (ev == nullptr) {
+      if (!in_list(free_, idx)) free_.push_back(idx);
+    } else if (!in_pending(idx)) {
+      pending_.emplace_back(idx, ev);
+    }
+    cv_.notify_all();
+  }
+
+  bool manages(const void* ptr) const {
+    std::lock_guard<std::mutex> lk(mu_);
+    if (!slab_.defined()) return false;
+    const uint8_t* base = slab_.data_ptr<uint8_t>();
+    const auto* p = static_cast<const uint8_t*>(ptr);
+    return p >= base && p < base + total_slots_ * slot_bytes_;
+  }
+
+  std::map<std::string, double> stats() const {
+    std::lock_guard<std::mutex> lk(mu_);
+    std::map<std::string, double> out;
+    out["slab_slots"] = (double)total_slots_;
+    out["slab_slot_bytes"] = (double)slot_bytes_;
+    out["slab_free"] = (double)free_.size();
+    out["slab_pending"] = (double)pending_.size();
+    out["slab_backpressure_waits"] = (double)backpressure_waits_.load();
+    return out;
+  }
+
+ private:
+  static bool in_list(const std::vector<int64_t>& v, int64_t x) {
+    for (auto e : v) {
+      if (e == x) return true;
+    }
+    return false;
+  }
+  bool in_pending(int64_t idx) const {
+    for (const auto& p : pending_) {
+      if (p.first == idx) return true;
+    }
+    return false;
+  }
+  void reap_pending() {
+    for (auto it = pending_.begin(); it != pending_.end();) {
+      if (it->second->query()) {
+        free_.push_back(it->first);
+        it = pending_.erase(it);
+      } else {
+        ++it;
+      }
+    }
+  }
+
+  const int64_t budget_bytes_;
+  const int64_t min_slots_;
+  mutable std::mutex mu_;
+  std::condition_variable cv_;
+  torch::Tensor slab_;
+  int64_t slot_bytes_ = 0;
+  int64_t total_slots_ = 0;
+  std::vector<int64_t> free_;
+  std::vector<std::pair<int64_t, std::shared_ptr<at::cuda::CUDAEvent>>>
+      pending_;
+  std::atomic<int64_t> backpressure_waits_{0};
+};
 
 inline TensorNest batch_nests(const std::vector<const TensorNest*>& nests,
                               int64_t batch_dim) {
@@ -283,17 +435,32 @@ class BatchingQueue {
         at::cuda::CUDAStreamGuard guard(*copy_stream_);
         batched = batch_nests_to_device(ptrs, batch_dim_, *output_device_);
       }
-      at::cuda::CUDAEvent ev;
-      ev.record(*copy_stream_);
-      ev.block(current);
+      auto ev = std::make_shared<at::cuda::CUDAEvent>();
+      ev->record(*copy_stream_);
+      ev->block(current);
       // The batch blocks were allocated on the copy stream but are consumed
       // (and eventually freed) on the caller's stream: tell the caching
       // allocator, or it may recycle them while the consumer still reads.
       batched.for_each([&current](const torch::Tensor& t) {
         t.record_stream(current);
       });
+      if (source_pool_) {
+        // Slab-pooled rollout slots recycle once the H2D copies complete.
+        for (const auto* n : ptrs) {
+          n->for_each([this, &ev](const torch::Tensor& t) {
+            source_pool_->mark_consumed(t.data_ptr(), ev);
+          });
+        }
+      }
     } else {
       batched = batch_nests(ptrs, batch_dim_);
+      if (source_pool_) {
+        for (const auto* n : ptrs) {
+          n->for_each([this](const torch::Tensor& t) {
+            source_pool_->mark_consumed(t.data_ptr(), nullptr);
+          });
+        }
+      }
     }
     auto result = std::make_pair(std::move(batched),
                                  static_cast<int64_t>(items.size()));
@@ -313,7 +480,14 @@ class BatchingQueue {
       out["avg_wait_ms"] = wait_stats_.sum_us / 1e3 / wait_stats_.n;
       out["avg_cat_ms"] = cat_stats_.sum_us / 1e3 / cat_stats_.n;
     }
+    if (source_pool_) {
+      for (const auto& kv : source_pool_->stats()) out[kv.first] = kv.second;
+    }
     return out;
+  }
+
+  void set_source_pool(std::shared_ptr<PinnedSlabPool> pool) {
+    source_pool_ = std::move(pool);
   }
 
  private:
@@ -324,6 +498,7 @@ class BatchingQueue {
   const std::optional<std::chrono::milliseconds> timeout_;
   const bool check_inputs_;
   BoundedQueue<TensorNest> queue_;
+  std::shared_ptr<PinnedSlabPool> source_pool_;
   std::unique_ptr<at::cuda::CUDAStream> copy_stream_;
   mutable StageStats wait_stats_;
   mutable StageStats cat_stats_;
