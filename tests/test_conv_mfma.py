@@ -18,9 +18,9 @@ else:  # pragma: no cover
     pytest.skip("requires ROCm GPU", allow_module_level=True)
 
 
-def _make_convs(seed=0, device="cuda"):
+def _make_convs(seed=0, device="cuda", in_ch=4):
     torch.manual_seed(seed)
-    c1 = torch.nn.Conv2d(4, 32, 8, stride=4).to(device)
+    c1 = torch.nn.Conv2d(in_ch, 32, 8, stride=4).to(device)
     c2 = torch.nn.Conv2d(32, 64, 4, stride=2).to(device)
     c3 = torch.nn.Conv2d(64, 64, 3, stride=1).to(device)
     return c1, c2, c3
@@ -34,11 +34,13 @@ def _eager_trunk(frames, c1, c2, c3):
     return x.reshape(frames.shape[0], -1)
 
 
-@pytest.mark.parametrize("N", [1, 7, 32, 500])
-def test_trunk_fwd_matches_eager(N):
-    c1, c2, c3 = _make_convs()
+@pytest.mark.parametrize("N,shape", [(1, (4, 84, 84)), (7, (4, 84, 84)),
+                                     (32, (4, 84, 84)), (500, (4, 84, 84)),
+                                     (3, (3, 210, 160)), (40, (3, 210, 160))])
+def test_trunk_fwd_matches_eager(N, shape):
+    c1, c2, c3 = _make_convs(in_ch=shape[0])
     torch.manual_seed(N)
-    frames = torch.randint(0, 256, (N, 4, 84, 84), dtype=torch.uint8,
+    frames = torch.randint(0, 256, (N, *shape), dtype=torch.uint8,
                            device="cuda")
     ext = ops_mod.require_ext()
     w1p, w2p, w3p = tbf._pack_trunk_weights(c1.weight, c2.weight, c3.weight)
@@ -67,7 +69,9 @@ def _semi_oracle_grads(frames, d_out, c1, c2, c3, a1, a2, out3):
     a1f = a1.permute(0, 3, 1, 2).float()  # NHWC bf16 -> NCHW fp32
     a2f = a2.permute(0, 3, 1, 2).float()
     N = frames.shape[0]
-    d3 = (d_out * (out3 > 0)).view(N, 64, 7, 7)
+    oh3 = a2.shape[1] - 2
+    ow3 = a2.shape[2] - 2
+    d3 = (d_out * (out3 > 0)).view(N, 64, oh3, ow3)
     d3 = _bf(d3)
     dw3 = conv2d_weight(_bf(a2f), c3.weight.shape, d3)
     db3 = d3.sum((0, 2, 3))
@@ -83,13 +87,16 @@ def _semi_oracle_grads(frames, d_out, c1, c2, c3, a1, a2, out3):
     return dw1, db1, dw2, db2, dw3, db3
 
 
-@pytest.mark.parametrize("N", [5, 96])
-def test_trunk_backward_matches_eager(N):
-    c1, c2, c3 = _make_convs(seed=3)
+@pytest.mark.parametrize("N,shape", [(5, (4, 84, 84)), (96, (4, 84, 84)),
+                                     (6, (3, 210, 160))])
+def test_trunk_backward_matches_eager(N, shape):
+    c1, c2, c3 = _make_convs(seed=3, in_ch=shape[0])
     torch.manual_seed(N + 100)
-    frames = torch.randint(0, 256, (N, 4, 84, 84), dtype=torch.uint8,
+    frames = torch.randint(0, 256, (N, *shape), dtype=torch.uint8,
                            device="cuda")
-    d_out = torch.randn(N, 3136, device="cuda")
+    with torch.no_grad():
+        nfeat = _eager_trunk(frames[:1], c1, c2, c3).shape[1]
+    d_out = torch.randn(N, nfeat, device="cuda")
 
     # Kernel path (grab the stash for the semi-oracle).
     ext = ops_mod.require_ext()
@@ -113,7 +120,7 @@ def test_trunk_backward_matches_eager(N):
         assert err < 1e-2, f"{name}: rel-max err {err:.4f}"
 
     # Against the true fp32 oracle the gradient direction must still agree.
-    c1e, c2e, c3e = _make_convs(seed=3)
+    c1e, c2e, c3e = _make_convs(seed=3, in_ch=shape[0])
     ref = _eager_trunk(frames, c1e, c2e, c3e)
     ref.backward(d_out)
     for ours, theirs, name in [

@@ -64,13 +64,16 @@ def test_shallow_runner_matches_model(use_lstm):
     (action, logits, baseline, new_state), ref_out, ref_state = _serve_once(
         model, b=5, use_lstm=use_lstm
     )
-    torch.testing.assert_close(logits, ref_out[1].cpu(), rtol=1e-4, atol=1e-4)
-    torch.testing.assert_close(baseline, ref_out[2].cpu(), rtol=1e-4,
-                               atol=1e-4)
+    # The runner serves the trunk from the bf16 MFMA kernels while the
+    # reference model forward here runs fp32; tolerances are set for that
+    # operand rounding (behavior-policy logits, not learner math).
+    torch.testing.assert_close(logits, ref_out[1].cpu(), rtol=5e-2, atol=5e-3)
+    torch.testing.assert_close(baseline, ref_out[2].cpu(), rtol=5e-2,
+                               atol=5e-3)
     assert action.shape == (1, 5) and action.dtype == torch.int64
     if use_lstm:
         for s, r in zip(new_state, ref_state):
-            torch.testing.assert_close(s, r.cpu(), rtol=1e-4, atol=1e-4)
+            torch.testing.assert_close(s, r.cpu(), rtol=5e-2, atol=5e-3)
 
 
 def test_deep_runner_matches_model():

@@ -303,11 +303,23 @@ class _AtariTrunkMfma(torch.autograd.Function):
         dw3p, db3 = ext.conv_trunk_wgrad3(a2, d3m)
         dw2p, db2 = ext.conv_trunk_wgrad2(a1, d2)
         dw1p, db1 = ext.conv_trunk_wgrad1(frames, d1)
-        # Kernel dW layout is [ky][co][kx*ci+c] ([ky][co][c*kw+kx] for conv1);
-        # permute back to PyTorch [co][ci][kh][kw].
-        dw3 = dw3p.view(3, 64, 3, 64).permute(1, 3, 0, 2).contiguous()
-        dw2 = dw2p.view(4, 64, 4, 32).permute(1, 3, 0, 2).contiguous()
-        dw1 = dw1p.view(8, 32, 4, 8).permute(1, 2, 0, 3).contiguous()
+
+        # Kernel dW layout is [ky][co][kx*ci+c] ([ky][co][c*kw+kx] for
+        # conv1, K-columns padded to 16); permute back to [co][ci][kh][kw].
+        def unpack(dwp, wshape, c_major):
+            co, ci, kh, kw = wshape
+            kwcp = dwp.shape[-1]
+            dwp = dwp.view(kh, co, kwcp)[..., :kw * ci]
+            if c_major:  # conv1: (c, kx) minor order
+                return (dwp.reshape(kh, co, ci, kw)
+                        .permute(1, 2, 0, 3).contiguous())
+            return (dwp.reshape(kh, co, kw, ci)
+                    .permute(1, 3, 0, 2).contiguous())
+
+        w1_shape = (d1.shape[-1], frames.shape[1], 8, 8)
+        dw3 = unpack(dw3p, w3.shape, c_major=False)
+        dw2 = unpack(dw2p, w2.shape, c_major=False)
+        dw1 = unpack(dw1p, w1_shape, c_major=True)
         return None, dw1, db1, dw2, db2, dw3, db3
 
 
@@ -328,7 +340,8 @@ def atari_trunk(frames, conv1, conv2, conv3):
     force = os.environ.get("TBAMD_TRUNK")  # mfma | valu | lib (benchmarking)
     if force == "lib":
         return None
-    mfma_ok = tuple(frames.shape[1:]) == (4, 84, 84) and force != "valu"
+    mfma_ok = tuple(frames.shape[1:]) in ((4, 84, 84), (3, 210, 160)) \
+        and force != "valu"
     grad = torch.is_grad_enabled() and (
         conv1.weight.requires_grad or frames.requires_grad
     )

@@ -69,7 +69,7 @@ __device__ __forceinline__ void lds_write8_swz(char* smem, int elem, bf16x8 v) {
 // ---------------------------------------------------------------------------
 
 template <int CI, int IH, int IW, int KH, int KW, int ST, int CO, int OH,
-          int OW>
+          int OW, int OYT = OH>
 __global__ __launch_bounds__(kThreads) void conv1_u8_kernel(
     const uint8_t* __restrict__ in,  // [N, CI, IH, IW]
     const __bf16* __restrict__ W,    // [CO, CI*KH*KW] (c,ky,kx)-major
@@ -78,37 +78,43 @@ __global__ __launch_bounds__(kThreads) void conv1_u8_kernel(
     int N) {
   constexpr int IWP = (IW + 7) & ~7;  // pad rows to 8 elements (16 B)
   constexpr int K = CI * KH * KW;
-  constexpr int M_BLK = OH * OW;
+  constexpr int LROWS = (OYT - 1) * ST + KH;  // staged input rows per band
+  constexpr int M_BLK = OYT * OW;
   constexpr int MF = (M_BLK + 15) / 16;
   constexpr int NF = CO / 16;
   constexpr int MAX_MF = (MF + kWaves - 1) / kWaves;
+  constexpr int BANDS = (OH + OYT - 1) / OYT;
   static_assert(KW == 8, "conv1 kernel assumes KW == 8 fragment runs");
 
-  extern __shared__ char smem[];  // CI * IH * IWP bf16
+  extern __shared__ char smem[];  // CI * LROWS * IWP bf16
   __bf16* img = reinterpret_cast<__bf16*>(smem);
 
-  const int s = blockIdx.x;
+  const int s = blockIdx.x / BANDS;
+  const int band = blockIdx.x % BANDS;
+  const int oy0 = band * OYT;
+  const int ly0 = oy0 * ST;
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
   const int ln = lane & 15;
   const int lg = lane >> 4;
 
-  // Stage the whole sample, converting u8 -> bf16/255. One thread per
-  // 16-px chunk (rows are 84 px = 5 full chunks + a 4-px tail).
+  // Stage this band's input rows, converting u8 -> bf16/255. One thread
+  // per 16-px chunk (84-px rows = 5 full chunks + a 4-px tail).
   {
     constexpr int CHUNKS_PER_ROW = (IW + 15) / 16;
-    constexpr int NCHUNK = CI * IH * CHUNKS_PER_ROW;
+    constexpr int NCHUNK = CI * LROWS * CHUNKS_PER_ROW;
     const uint8_t* base = in + (int64_t)s * CI * IH * IW;
     for (int idx = tid; idx < NCHUNK; idx += kThreads) {
-      const int c = idx / (IH * CHUNKS_PER_ROW);
-      const int rem = idx % (IH * CHUNKS_PER_ROW);
-      const int row = rem / CHUNKS_PER_ROW;
+      const int c = idx / (LROWS * CHUNKS_PER_ROW);
+      const int rem = idx % (LROWS * CHUNKS_PER_ROW);
+      const int lrow = rem / CHUNKS_PER_ROW;
+      const int row = ly0 + lrow;
       const int ch = rem % CHUNKS_PER_ROW;
       const int x0 = ch * 16;
-      const int npx = min(16, IW - x0);
+      const int npx = (row < IH) ? min(16, IW - x0) : 0;
       const uint8_t* src = base + (c * IH + row) * IW + x0;
-      __bf16* dst = &img[(c * IH + row) * IWP + x0];
+      __bf16* dst = &img[(c * LROWS + lrow) * IWP + x0];
       constexpr float kInv = 1.0f / 255.0f;
       if (npx == 16) {
         // Row starts are only 4 B-aligned (84 px rows): four u32 loads.
@@ -130,7 +136,7 @@ __global__ __launch_bounds__(kThreads) void conv1_u8_kernel(
       }
     }
     // Zero the row pad (A-fragments of the right edge read into it).
-    for (int idx = tid; idx < CI * IH * (IWP - IW); idx += kThreads) {
+    for (int idx = tid; idx < CI * LROWS * (IWP - IW); idx += kThreads) {
       const int r = idx / (IWP - IW);
       img[r * IWP + IW + idx % (IWP - IW)] = (__bf16)0.f;
     }
@@ -138,13 +144,14 @@ __global__ __launch_bounds__(kThreads) void conv1_u8_kernel(
   __syncthreads();
 
   // Per-wave fragment bookkeeping.
-  int aoff[MAX_MF];  // LDS element offset of (oy*ST) row, +ox*ST
+  int aoff[MAX_MF];  // LDS element offset of the band-local (oy*ST) row
   int nmf = 0;
   for (int f = wave; f < MF; f += kWaves, ++nmf) {
     const int m = min(f * 16 + ln, M_BLK - 1);
     const int oy = m / OW, ox = m % OW;
     aoff[nmf] = (oy * ST) * IWP + ox * ST;
   }
+  const int mvalid = min(OYT, OH - oy0) * OW;
 
   f32x4 acc[MAX_MF][NF];
 #pragma unroll
@@ -163,7 +170,7 @@ __global__ __launch_bounds__(kThreads) void conv1_u8_kernel(
     for (int j = 0; j < NF; ++j) {
       bfr[j] = *reinterpret_cast<const bf16x8*>(&W[(j * 16 + ln) * K + k]);
     }
-    const int rowoff = (c * IH + ky) * IWP;
+    const int rowoff = (c * LROWS + ky) * IWP;
 #pragma unroll
     for (int f = 0; f < MAX_MF; ++f) {
       if (f >= nmf) break;
@@ -185,7 +192,7 @@ __global__ __launch_bounds__(kThreads) void conv1_u8_kernel(
   }
 
   // Epilogue: +bias, ReLU, NHWC bf16.
-  __bf16* obase = out + (int64_t)s * OH * OW * CO;
+  __bf16* obase = out + ((int64_t)s * OH + oy0) * OW * CO;
   int fi = 0;
   for (int f = wave; f < MF; f += kWaves, ++fi) {
 #pragma unroll
@@ -195,9 +202,9 @@ __global__ __launch_bounds__(kThreads) void conv1_u8_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int m = f * 16 + lg * 4 + r;
-        if (m >= M_BLK) continue;
+        if (m >= mvalid) continue;
         const float v = acc[fi][j][r] + b;
-        obase[m * CO + cch] = (__bf16)(v > 0.f ? v : 0.f);
+        obase[(int64_t)m * CO + cch] = (__bf16)(v > 0.f ? v : 0.f);
       }
     }
   }
@@ -396,22 +403,24 @@ __global__ void mask_d3_kernel(const float* __restrict__ dflat,
 // ---------------------------------------------------------------------------
 
 template <int CI, int XH, int XW, int KH, int KW, int ST, int CO, int OH,
-          int OW, int MC, bool X_U8>
+          int OW, int MC, bool X_U8,
+          int KWCP = ((KW * CI + 15) / 16) * 16>
 __global__ __launch_bounds__(kThreads) void wgrad_kernel(
     const void* __restrict__ xin,    // NHWC bf16 [N,XH,XW,CI] or u8 NCHW
     const __bf16* __restrict__ dy,   // [N, OH, OW, CO]
-    float* __restrict__ partials,    // [nchunks, KH, CO, KW*CI]
+    float* __restrict__ partials,    // [nchunks, KH, CO, KWCP]
     float* __restrict__ db_partials,  // [nchunks, CO]
     int N) {
-  constexpr int KWC = KW * CI;
+  constexpr int KWC = KW * CI;  // real K-columns; KWCP pads to MFMA tiles
   constexpr int MPAD = 40;  // 32 + 8: pads the transposed tiles' rows
-  constexpr int OF = (CO / 16) * (KWC / 16);
+  constexpr int OF = (CO / 16) * (KWCP / 16);
   constexpr int PER_WAVE = (OF + kWaves - 1) / kWaves;
-  static_assert(KWC % 16 == 0, "KW*CI must tile by 16");
+  static_assert(KWC % 8 == 0, "KW*CI must be whole 16 B chunks");
+  static_assert(KWCP % 16 == 0 && KWCP >= KWC, "bad KWCP");
   static_assert(CO % 16 == 0, "CO must tile by 16");
 
   __shared__ __bf16 sDYT[CO][MPAD];
-  __shared__ __bf16 sXT[KWC][MPAD];
+  __shared__ __bf16 sXT[KWCP][MPAD];
 
   const int ky = blockIdx.x;
   const int chunk = blockIdx.y;
@@ -427,6 +436,11 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
 #pragma unroll
   for (int i = 0; i < PER_WAVE; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
   float db = 0.f;
+  // Zero the K-column pad rows once (staging never writes them).
+  for (int i = tid; i < (KWCP - KWC) * MPAD; i += kThreads) {
+    sXT[KWC + i / MPAD][i % MPAD] = (__bf16)0.f;
+  }
+  if (KWCP != KWC) __syncthreads();
 
   const int64_t mend = (m0 + MC < M) ? m0 + MC : M;
   for (int64_t ms = m0; ms < mend; ms += 32) {
@@ -495,8 +509,8 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
     for (int i = 0; i < PER_WAVE; ++i) {
       const int of = wave * PER_WAVE + i;
       if (of >= OF) break;
-      const int ni = of / (KWC / 16);
-      const int ki = of % (KWC / 16);
+      const int ni = of / (KWCP / 16);
+      const int ki = of % (KWCP / 16);
       const bf16x8 a =
           *reinterpret_cast<const bf16x8*>(&sDYT[ni * 16 + ln][8 * lg]);
       const bf16x8 b =
@@ -518,16 +532,16 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
 
   // ---- write fp32 partials ----
   float* pbase =
-      partials + ((int64_t)chunk * KH + ky) * CO * KWC;
+      partials + ((int64_t)chunk * KH + ky) * CO * KWCP;
 #pragma unroll
   for (int i = 0; i < PER_WAVE; ++i) {
     const int of = wave * PER_WAVE + i;
     if (of >= OF) break;
-    const int ni = of / (KWC / 16);
-    const int ki = of % (KWC / 16);
+    const int ni = of / (KWCP / 16);
+    const int ki = of % (KWCP / 16);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      pbase[(ni * 16 + lg * 4 + r) * KWC + ki * 16 + ln] = acc[i][r];
+      pbase[(ni * 16 + lg * 4 + r) * KWCP + ki * 16 + ln] = acc[i][r];
     }
   }
   if (ky == 0 && wave == 0 && lane < CO) {
@@ -554,7 +568,7 @@ __global__ void reduce_partials_kernel(const float* __restrict__ partials,
 
 struct G1 {  // conv1 8x8 s4, 4->32: 84x84 -> 20x20
   static constexpr int CI = 4, IH = 84, IW = 84, KH = 8, KW = 8, ST = 4;
-  static constexpr int CO = 32, OH = 20, OW = 20;
+  static constexpr int CO = 32, OH = 20, OW = 20, OYT = 20;
   static constexpr int IWP = 88;
   static constexpr size_t LDS = (size_t)CI * IH * IWP * 2;
 };
@@ -583,6 +597,48 @@ struct D2 {  // dgrad conv2: dY [9,9,64] dil 2 pad 3 -> dX [20,20,32]
   static constexpr size_t LDS = (size_t)SB * ((OYT - 1) * ST + KH) * IWL * CI * 2;
 };
 
+// Full-resolution (210x160x3) geometry, BASELINE config 5. Output-row
+// banding keeps every LDS tile under 2-blocks/CU budget.
+struct F1 {  // conv1 8x8 s4, 3->32: 210x160 -> 51x39
+  static constexpr int CI = 3, IH = 210, IW = 160, KH = 8, KW = 8, ST = 4;
+  static constexpr int CO = 32, OH = 51, OW = 39, OYT = 13;
+  static constexpr int IWP = 160;
+  static constexpr size_t LDS =
+      (size_t)CI * ((OYT - 1) * ST + KH) * IWP * 2;
+};
+struct F2 {  // conv2 4x4 s2, 32->64: 51x39 -> 24x18
+  static constexpr int CI = 32, IH = 51, IW = 39, KH = 4, KW = 4, ST = 2;
+  static constexpr int CO = 64, OH = 24, OW = 18, SB = 1, OYT = 12;
+  static constexpr size_t LDS =
+      (size_t)((OYT - 1) * ST + KH) * IW * CI * 2;
+};
+struct F3 {  // conv3 3x3 s1, 64->64: 24x18 -> 22x16 (flat 22528)
+  static constexpr int CI = 64, IH = 24, IW = 18, KH = 3, KW = 3, ST = 1;
+  static constexpr int CO = 64, OH = 22, OW = 16, SB = 1, OYT = 22;
+  static constexpr size_t LDS =
+      (size_t)((OYT - 1) * ST + KH) * IW * CI * 2;
+};
+struct FD3 {  // dgrad conv3 FR: dY [22,16,64] pad 2 -> dX [24,18,64]
+  static constexpr int CI = 64, IHR = 22, IWR = 16, PAD = 2, DIL = 1;
+  static constexpr int KH = 3, KW = 3, ST = 1, CO = 64, OH = 24, OW = 18;
+  static constexpr int SB = 1, OYT = 12;
+  static constexpr int IWL = DIL * (IWR - 1) + 1 + 2 * PAD;
+  static constexpr size_t LDS =
+      (size_t)((OYT - 1) * ST + KH) * IWL * CI * 2;
+};
+struct FD2 {  // dgrad conv2 FR: dY [24,18,64] dil 2 pad 3 -> dX [51,39,32]
+  static constexpr int CI = 64, IHR = 24, IWR = 18, PAD = 3, DIL = 2;
+  static constexpr int KH = 4, KW = 4, ST = 1, CO = 32, OH = 51, OW = 39;
+  static constexpr int SB = 1, OYT = 8;
+  static constexpr int IWL = DIL * (IWR - 1) + 1 + 2 * PAD;
+  static constexpr size_t LDS =
+      (size_t)((OYT - 1) * ST + KH) * IWL * CI * 2;
+};
+
+bool is_geom(const torch::Tensor& frames, int C, int H, int W) {
+  return frames.size(1) == C && frames.size(2) == H && frames.size(3) == W;
+}
+
 void check_bf16(const torch::Tensor& t, const char* what) {
   TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16 &&
                   t.is_contiguous(),
@@ -590,6 +646,81 @@ void check_bf16(const torch::Tensor& t, const char* what) {
 }
 
 }  // namespace
+
+template <typename C1, typename C2, typename C3>
+std::vector<torch::Tensor> trunk_fwd_impl(torch::Tensor frames,
+                                          torch::Tensor w1, torch::Tensor b1,
+                                          torch::Tensor w2, torch::Tensor b2,
+                                          torch::Tensor w3, torch::Tensor b3,
+                                          bool want_stash) {
+  check_bf16(w1, "w1");
+  check_bf16(w2, "w2");
+  check_bf16(w3, "w3");
+  auto fr = frames.contiguous();
+  const int N = fr.size(0);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  auto bfopt = w1.options();
+
+  auto a1 = torch::empty({N, C1::OH, C1::OW, C1::CO}, bfopt);
+  auto a2 = torch::empty({N, C2::OH, C2::OW, C2::CO}, bfopt);
+  auto out3 = torch::empty({N, (int64_t)C3::CO * C3::OH * C3::OW},
+                           bfopt.dtype(torch::kFloat32));
+  if (N == 0) return {out3, a1, a2};
+
+  constexpr int kBands1 = (C1::OH + C1::OYT - 1) / C1::OYT;
+  hipLaunchKernelGGL(
+      (conv1_u8_kernel<C1::CI, C1::IH, C1::IW, C1::KH, C1::KW, C1::ST,
+                       C1::CO, C1::OH, C1::OW, C1::OYT>),
+      dim3(N * kBands1), dim3(kThreads), C1::LDS, stream,
+      fr.data_ptr<uint8_t>(),
+      reinterpret_cast<const __bf16*>(w1.data_ptr()), b1.data_ptr<float>(),
+      reinterpret_cast<__bf16*>(a1.data_ptr()), N);
+
+  // Small (inference-sized) batches need grid = N x bands for occupancy
+  // (measured: conv3 at batch ~100 was the top GPU kernel purely from
+  // sample-group underfill).
+  if (N < 1024 && C2::SB > 1) {
+    constexpr size_t kLds2 = (size_t)C2::IH * C2::IW * C2::CI * 2;
+    hipLaunchKernelGGL(
+        (conv_nhwc_kernel<C2::CI, C2::IH, C2::IW, 0, 1, C2::KH, C2::KW,
+                          C2::ST, C2::CO, C2::OH, C2::OW, 1, C2::OH, 0, 384>),
+        dim3(N), dim3(384), kLds2, stream,
+        reinterpret_cast<const __bf16*>(a1.data_ptr()),
+        reinterpret_cast<const __bf16*>(w2.data_ptr()), b2.data_ptr<float>(),
+        nullptr, a2.data_ptr(), N);
+    constexpr size_t kLds3 = (size_t)C3::IH * C3::IW * C3::CI * 2;
+    hipLaunchKernelGGL(
+        (conv_nhwc_kernel<C3::CI, C3::IH, C3::IW, 0, 1, C3::KH, C3::KW,
+                          C3::ST, C3::CO, C3::OH, C3::OW, 1, C3::OH, 1, 256>),
+        dim3(N), dim3(256), kLds3, stream,
+        reinterpret_cast<const __bf16*>(a2.data_ptr()),
+        reinterpret_cast<const __bf16*>(w3.data_ptr()), b3.data_ptr<float>(),
+        nullptr, out3.data_ptr(), N);
+    return want_stash ? std::vector<torch::Tensor>{out3, a1, a2}
+                      : std::vector<torch::Tensor>{out3};
+  }
+
+  constexpr int kBands2 = (C2::OH + C2::OYT - 1) / C2::OYT;
+  hipLaunchKernelGGL(
+      (conv_nhwc_kernel<C2::CI, C2::IH, C2::IW, 0, 1, C2::KH, C2::KW, C2::ST,
+                        C2::CO, C2::OH, C2::OW, C2::SB, C2::OYT, 0>),
+      dim3(((N + C2::SB - 1) / C2::SB) * kBands2), dim3(kThreads), C2::LDS,
+      stream, reinterpret_cast<const __bf16*>(a1.data_ptr()),
+      reinterpret_cast<const __bf16*>(w2.data_ptr()), b2.data_ptr<float>(),
+      nullptr, a2.data_ptr(), N);
+
+  constexpr int kBands3 = (C3::OH + C3::OYT - 1) / C3::OYT;
+  hipLaunchKernelGGL(
+      (conv_nhwc_kernel<C3::CI, C3::IH, C3::IW, 0, 1, C3::KH, C3::KW, C3::ST,
+                        C3::CO, C3::OH, C3::OW, C3::SB, C3::OYT, 1>),
+      dim3(((N + C3::SB - 1) / C3::SB) * kBands3), dim3(kThreads), C3::LDS,
+      stream, reinterpret_cast<const __bf16*>(a2.data_ptr()),
+      reinterpret_cast<const __bf16*>(w3.data_ptr()), b3.data_ptr<float>(),
+      nullptr, out3.data_ptr(), N);
+
+  if (!want_stash) return {out3};
+  return {out3, a1, a2};
+}
 
 std::vector<torch::Tensor> conv_trunk_fwd(torch::Tensor frames,
                                           torch::Tensor w1, torch::Tensor b1,
@@ -599,89 +730,57 @@ std::vector<torch::Tensor> conv_trunk_fwd(torch::Tensor frames,
   TORCH_CHECK(frames.is_cuda() && frames.scalar_type() == torch::kUInt8 &&
                   frames.dim() == 4,
               "conv_trunk_fwd: u8 [N,C,H,W] GPU frames required");
-  TORCH_CHECK(frames.size(1) == G1::CI && frames.size(2) == G1::IH &&
-                  frames.size(3) == G1::IW,
-              "conv_trunk_fwd: geometry is 4x84x84 (got ", frames.sizes(), ")");
-  check_bf16(w1, "w1");
-  check_bf16(w2, "w2");
-  check_bf16(w3, "w3");
-  auto fr = frames.contiguous();
-  const int N = fr.size(0);
-  auto stream = at::cuda::getCurrentCUDAStream();
-  auto bfopt = w1.options();
-
-  auto a1 = torch::empty({N, G1::OH, G1::OW, G1::CO}, bfopt);
-  auto a2 = torch::empty({N, G2::OH, G2::OW, G2::CO}, bfopt);
-  auto out3 = torch::empty({N, (int64_t)G3::CO * G3::OH * G3::OW},
-                           bfopt.dtype(torch::kFloat32));
-  if (N == 0) return {out3, a1, a2};
-
-  hipLaunchKernelGGL(
-      (conv1_u8_kernel<G1::CI, G1::IH, G1::IW, G1::KH, G1::KW, G1::ST, G1::CO,
-                       G1::OH, G1::OW>),
-      dim3(N), dim3(kThreads), G1::LDS, stream, fr.data_ptr<uint8_t>(),
-      reinterpret_cast<const __bf16*>(w1.data_ptr()), b1.data_ptr<float>(),
-      reinterpret_cast<__bf16*>(a1.data_ptr()), N);
-
-  // Small (inference-sized) batches need grid = N for occupancy: the
-  // sample-grouped variants would launch only N/SB blocks (measured: conv3
-  // at batch ~100 was the top GPU kernel purely from underfill).
-  if (N < 1024) {
-    constexpr size_t kLds2 = (size_t)G2::IH * G2::IW * G2::CI * 2;
-    hipLaunchKernelGGL(
-        (conv_nhwc_kernel<G2::CI, G2::IH, G2::IW, 0, 1, G2::KH, G2::KW,
-                          G2::ST, G2::CO, G2::OH, G2::OW, 1, G2::OH, 0, 384>),
-        dim3(N), dim3(384), kLds2, stream,
-        reinterpret_cast<const __bf16*>(a1.data_ptr()),
-        reinterpret_cast<const __bf16*>(w2.data_ptr()), b2.data_ptr<float>(),
-        nullptr, a2.data_ptr(), N);
-    constexpr size_t kLds3 = (size_t)G3::IH * G3::IW * G3::CI * 2;
-    hipLaunchKernelGGL(
-        (conv_nhwc_kernel<G3::CI, G3::IH, G3::IW, 0, 1, G3::KH, G3::KW,
-                          G3::ST, G3::CO, G3::OH, G3::OW, 1, G3::OH, 1, 256>),
-        dim3(N), dim3(256), kLds3, stream,
-        reinterpret_cast<const __bf16*>(a2.data_ptr()),
-        reinterpret_cast<const __bf16*>(w3.data_ptr()), b3.data_ptr<float>(),
-        nullptr, out3.data_ptr(), N);
-    return want_stash ? std::vector<torch::Tensor>{out3, a1, a2}
-                      : std::vector<torch::Tensor>{out3};
+  if (is_geom(frames, 4, 84, 84)) {
+    return trunk_fwd_impl<G1, G2, G3>(frames, w1, b1, w2, b2, w3, b3,
+                                      want_stash);
   }
+  if (is_geom(frames, 3, 210, 160)) {
+    return trunk_fwd_impl<F1, F2, F3>(frames, w1, b1, w2, b2, w3, b3,
+                                      want_stash);
+  }
+  TORCH_CHECK(false, "conv_trunk_fwd: unsupported geometry ", frames.sizes());
+}
 
-  hipLaunchKernelGGL(
-      (conv_nhwc_kernel<G2::CI, G2::IH, G2::IW, 0, 1, G2::KH, G2::KW, G2::ST,
-                        G2::CO, G2::OH, G2::OW, G2::SB, G2::OYT, 0>),
-      dim3((N + G2::SB - 1) / G2::SB), dim3(kThreads), G2::LDS, stream,
-      reinterpret_cast<const __bf16*>(a1.data_ptr()),
-      reinterpret_cast<const __bf16*>(w2.data_ptr()), b2.data_ptr<float>(),
-      nullptr, a2.data_ptr(), N);
-
-  hipLaunchKernelGGL(
-      (conv_nhwc_kernel<G3::CI, G3::IH, G3::IW, 0, 1, G3::KH, G3::KW, G3::ST,
-                        G3::CO, G3::OH, G3::OW, G3::SB, G3::OYT, 1>),
-      dim3((N + G3::SB - 1) / G3::SB), dim3(kThreads), G3::LDS, stream,
-      reinterpret_cast<const __bf16*>(a2.data_ptr()),
-      reinterpret_cast<const __bf16*>(w3.data_ptr()), b3.data_ptr<float>(),
-      nullptr, out3.data_ptr(), N);
-
-  if (!want_stash) return {out3};
-  return {out3, a1, a2};
+template <typename C3>
+torch::Tensor mask_d3_impl(torch::Tensor d_out3, torch::Tensor out3) {
+  auto d = d_out3.contiguous();
+  auto o = out3.contiguous();
+  const int64_t N = d.size(0);
+  auto d3m = torch::empty({N, C3::OH, C3::OW, C3::CO},
+                          d.options().dtype(torch::kBFloat16));
+  const int64_t total = N * C3::CO * C3::OH * C3::OW;
+  const int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL((mask_d3_kernel<C3::CO, C3::OH, C3::OW>), dim3(blocks),
+                     dim3(256), 0, at::cuda::getCurrentCUDAStream(),
+                     d.data_ptr<float>(), o.data_ptr<float>(),
+                     reinterpret_cast<__bf16*>(d3m.data_ptr()), total);
+  return d3m;
 }
 
 torch::Tensor conv_trunk_mask_d3(torch::Tensor d_out3, torch::Tensor out3) {
   TORCH_CHECK(d_out3.is_cuda() && d_out3.scalar_type() == torch::kFloat32,
               "mask_d3: fp32 grad required");
-  auto d = d_out3.contiguous();
-  auto o = out3.contiguous();
-  const int64_t N = d.size(0);
-  auto d3m = torch::empty({N, G3::OH, G3::OW, G3::CO},
-                          d.options().dtype(torch::kBFloat16));
-  const int64_t total = N * G3::CO * G3::OH * G3::OW;
-  const int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
-  hipLaunchKernelGGL((mask_d3_kernel<G3::CO, G3::OH, G3::OW>), dim3(blocks),
-                     dim3(256), 0, at::cuda::getCurrentCUDAStream(),
-                     d.data_ptr<float>(), o.data_ptr<float>(),
-                     reinterpret_cast<__bf16*>(d3m.data_ptr()), total);
-  return d3m;
+  const int64_t per = d_out3.size(1);
+  if (per == G3::CO * G3::OH * G3::OW) return mask_d3_impl<G3>(d_out3, out3);
+  if (per == F3::CO * F3::OH * F3::OW) return mask_d3_impl<F3>(d_out3, out3);
+  TORCH_CHECK(false, "mask_d3: unsupported feature size ", per);
+}
+
+template <typename D>
+torch::Tensor dgrad_impl(torch::Tensor dy, torch::Tensor wr,
+                         torch::Tensor act) {
+  const int N = dy.size(0);
+  auto dx = torch::empty_like(act);
+  constexpr int kBands = (D::OH + D::OYT - 1) / D::OYT;
+  hipLaunchKernelGGL(
+      (conv_nhwc_kernel<D::CI, D::IHR, D::IWR, D::PAD, D::DIL, D::KH, D::KW,
+                        D::ST, D::CO, D::OH, D::OW, D::SB, D::OYT, 2>),
+      dim3(((N + D::SB - 1) / D::SB) * kBands), dim3(kThreads), D::LDS,
+      at::cuda::getCurrentCUDAStream(),
+      reinterpret_cast<const __bf16*>(dy.data_ptr()),
+      reinterpret_cast<const __bf16*>(wr.data_ptr()), nullptr,
+      reinterpret_cast<const __bf16*>(act.data_ptr()), dx.data_ptr(), N);
+  return dx;
 }
 
 torch::Tensor conv_trunk_dgrad3(torch::Tensor d3m, torch::Tensor w3r,
@@ -689,18 +788,13 @@ torch::Tensor conv_trunk_dgrad3(torch::Tensor d3m, torch::Tensor w3r,
   check_bf16(d3m, "d3m");
   check_bf16(w3r, "w3r");
   check_bf16(a2, "a2");
-  const int N = d3m.size(0);
-  auto d2 = torch::empty_like(a2);
-  hipLaunchKernelGGL(
-      (conv_nhwc_kernel<D3::CI, D3::IHR, D3::IWR, D3::PAD, D3::DIL, D3::KH,
-                        D3::KW, D3::ST, D3::CO, D3::OH, D3::OW, D3::SB,
-                        D3::OYT, 2>),
-      dim3((N + D3::SB - 1) / D3::SB), dim3(kThreads), D3::LDS,
-      at::cuda::getCurrentCUDAStream(),
-      reinterpret_cast<const __bf16*>(d3m.data_ptr()),
-      reinterpret_cast<const __bf16*>(w3r.data_ptr()), nullptr,
-      reinterpret_cast<const __bf16*>(a2.data_ptr()), d2.data_ptr(), N);
-  return d2;
+  if (d3m.size(1) == D3::IHR && d3m.size(2) == D3::IWR) {
+    return dgrad_impl<D3>(d3m, w3r, a2);
+  }
+  if (d3m.size(1) == FD3::IHR && d3m.size(2) == FD3::IWR) {
+    return dgrad_impl<FD3>(d3m, w3r, a2);
+  }
+  TORCH_CHECK(false, "dgrad3: unsupported geometry ", d3m.sizes());
 }
 
 torch::Tensor conv_trunk_dgrad2(torch::Tensor d2, torch::Tensor w2r,
@@ -708,18 +802,13 @@ torch::Tensor conv_trunk_dgrad2(torch::Tensor d2, torch::Tensor w2r,
   check_bf16(d2, "d2");
   check_bf16(w2r, "w2r");
   check_bf16(a1, "a1");
-  const int N = d2.size(0);
-  auto d1 = torch::empty_like(a1);
-  hipLaunchKernelGGL(
-      (conv_nhwc_kernel<D2::CI, D2::IHR, D2::IWR, D2::PAD, D2::DIL, D2::KH,
-                        D2::KW, D2::ST, D2::CO, D2::OH, D2::OW, D2::SB,
-                        D2::OYT, 2>),
-      dim3((N + D2::SB - 1) / D2::SB), dim3(kThreads), D2::LDS,
-      at::cuda::getCurrentCUDAStream(),
-      reinterpret_cast<const __bf16*>(d2.data_ptr()),
-      reinterpret_cast<const __bf16*>(w2r.data_ptr()), nullptr,
-      reinterpret_cast<const __bf16*>(a1.data_ptr()), d1.data_ptr(), N);
-  return d1;
+  if (d2.size(1) == D2::IHR && d2.size(2) == D2::IWR) {
+    return dgrad_impl<D2>(d2, w2r, a1);
+  }
+  if (d2.size(1) == FD2::IHR && d2.size(2) == FD2::IWR) {
+    return dgrad_impl<FD2>(d2, w2r, a1);
+  }
+  TORCH_CHECK(false, "dgrad2: unsupported geometry ", d2.sizes());
 }
 
 namespace {
@@ -728,7 +817,7 @@ template <int CI, int XH, int XW, int KH, int KW, int ST, int CO, int OH,
           int OW, int MC, bool X_U8>
 std::vector<torch::Tensor> run_wgrad(const torch::Tensor& x,
                                      const torch::Tensor& dy) {
-  constexpr int KWC = KW * CI;
+  constexpr int KWC = ((KW * CI + 15) / 16) * 16;  // padded (kernel KWCP)
   const int N = dy.size(0);
   const int64_t M = (int64_t)N * OH * OW;
   const int nchunks = (int)((M + MC - 1) / MC);
@@ -766,24 +855,45 @@ std::vector<torch::Tensor> conv_trunk_wgrad1(torch::Tensor frames,
               "wgrad1: u8 frames");
   check_bf16(d1, "d1");
   auto fr = frames.contiguous();
-  return run_wgrad<G1::CI, G1::IH, G1::IW, G1::KH, G1::KW, G1::ST, G1::CO,
-                   G1::OH, G1::OW, 2048, true>(fr, d1);
+  if (is_geom(fr, 4, 84, 84)) {
+    return run_wgrad<G1::CI, G1::IH, G1::IW, G1::KH, G1::KW, G1::ST, G1::CO,
+                     G1::OH, G1::OW, 2048, true>(fr, d1);
+  }
+  if (is_geom(fr, 3, 210, 160)) {
+    return run_wgrad<F1::CI, F1::IH, F1::IW, F1::KH, F1::KW, F1::ST, F1::CO,
+                     F1::OH, F1::OW, 2048, true>(fr, d1);
+  }
+  TORCH_CHECK(false, "wgrad1: unsupported geometry ", fr.sizes());
 }
 
 std::vector<torch::Tensor> conv_trunk_wgrad2(torch::Tensor a1,
                                              torch::Tensor d2) {
   check_bf16(a1, "a1");
   check_bf16(d2, "d2");
-  return run_wgrad<G2::CI, G2::IH, G2::IW, G2::KH, G2::KW, G2::ST, G2::CO,
-                   G2::OH, G2::OW, 2048, false>(a1, d2);
+  if (a1.size(1) == G2::IH && a1.size(2) == G2::IW) {
+    return run_wgrad<G2::CI, G2::IH, G2::IW, G2::KH, G2::KW, G2::ST, G2::CO,
+                     G2::OH, G2::OW, 2048, false>(a1, d2);
+  }
+  if (a1.size(1) == F2::IH && a1.size(2) == F2::IW) {
+    return run_wgrad<F2::CI, F2::IH, F2::IW, F2::KH, F2::KW, F2::ST, F2::CO,
+                     F2::OH, F2::OW, 2048, false>(a1, d2);
+  }
+  TORCH_CHECK(false, "wgrad2: unsupported geometry ", a1.sizes());
 }
 
 std::vector<torch::Tensor> conv_trunk_wgrad3(torch::Tensor a2,
                                              torch::Tensor d3m) {
   check_bf16(a2, "a2");
   check_bf16(d3m, "d3m");
-  return run_wgrad<G3::CI, G3::IH, G3::IW, G3::KH, G3::KW, G3::ST, G3::CO,
-                   G3::OH, G3::OW, 1024, false>(a2, d3m);
+  if (a2.size(1) == G3::IH && a2.size(2) == G3::IW) {
+    return run_wgrad<G3::CI, G3::IH, G3::IW, G3::KH, G3::KW, G3::ST, G3::CO,
+                     G3::OH, G3::OW, 1024, false>(a2, d3m);
+  }
+  if (a2.size(1) == F3::IH && a2.size(2) == F3::IW) {
+    return run_wgrad<F3::CI, F3::IH, F3::IW, F3::KH, F3::KW, F3::ST, F3::CO,
+                     F3::OH, F3::OW, 1024, false>(a2, d3m);
+  }
+  TORCH_CHECK(false, "wgrad3: unsupported geometry ", a2.sizes());
 }
 
 }  // namespace tbamd
