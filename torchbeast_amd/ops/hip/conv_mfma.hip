@@ -232,8 +232,13 @@ __global__ __launch_bounds__(NT) void conv_nhwc_kernel(
     const float* __restrict__ bias,  // [CO] (MODE 0/1)
     const __bf16* __restrict__ act,  // [N, OH, OW, CO] (MODE 2)
     void* __restrict__ out, int N) {
-  constexpr int IHL = DIL * (IHR - 1) + 1 + 2 * PAD;
-  constexpr int IWL = DIL * (IWR - 1) + 1 + 2 * PAD;
+  // Logical tile sized from the OUTPUT window: when (input-kernel)%stride
+  // != 0 the forward drops a trailing row/column, so the transposed-conv
+  // (dgrad) windows reach one element past the dilated input — those
+  // positions are zeros via the staging guards.
+  constexpr int IWL0 = DIL * (IWR - 1) + 1 + 2 * PAD;
+  constexpr int IWL1 = (OW - 1) * ST + KW;
+  constexpr int IWL = IWL0 > IWL1 ? IWL0 : IWL1;
   constexpr int ROWE = IWL * CI;           // elements per logical row
   constexpr int LROWS = (OYT - 1) * ST + KH;  // staged logical rows
   constexpr int K = KH * KW * CI;
@@ -248,7 +253,6 @@ __global__ __launch_bounds__(NT) void conv_nhwc_kernel(
   static_assert(CI % 8 == 0 || (PAD == 0 && DIL == 1),
                 "pad/dilate staging assumes chunks within one x");
   static_assert(SB == 1 || OYT == OH, "multi-sample blocks stage full rows");
-  (void)IHL;
 
   extern __shared__ char smem[];  // SB * LROWS * ROWE bf16, swizzled
 
@@ -630,7 +634,9 @@ struct FD2 {  // dgrad conv2 FR: dY [24,18,64] dil 2 pad 3 -> dX [51,39,32]
   static constexpr int CI = 64, IHR = 24, IWR = 18, PAD = 3, DIL = 2;
   static constexpr int KH = 4, KW = 4, ST = 1, CO = 32, OH = 51, OW = 39;
   static constexpr int SB = 1, OYT = 8;
-  static constexpr int IWL = DIL * (IWR - 1) + 1 + 2 * PAD;
+  // Output-driven logical width (42): one past the dilated input, see the
+  // kernel's IWL derivation.
+  static constexpr int IWL = (OW - 1) * ST + KW;
   static constexpr size_t LDS =
       (size_t)((OYT - 1) * ST + KH) * IWL * CI * 2;
 };

@@ -222,7 +222,8 @@ class InferenceRunner {
         }
       }
       x = at::relu(x).reshape({bp, -1});
-    } else if (C == 4 && H == 84 && W == 84) {
+    } else if ((C == 4 && H == 84 && W == 84) ||
+               (C == 3 && H == 210 && W == 160)) {
       // MFMA implicit-GEMM trunk (bf16 operands, fp32 accumulate) — flat
       // per-batch cost at every dynamic batch size, unlike the per-sample
       // fused kernel whose grid is the batch (underfills the 256 CUs below
