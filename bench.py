@@ -176,10 +176,17 @@ def main():
         t.cpu() for t in model.initial_state(batch_size=1)
     )
     use_cpp_inference = use_cuda and not args.py_inference
-    # Measured on MI355X: the GPU-side slab gather loses ~12% to the
-    # pinned-cat + SDMA copy path (kernel reads over the host link are
-    # slower than the DMA engine). Opt-in for experiments.
-    use_obs_slab = use_cpp_inference and bool(os.environ.get("TBAMD_OBS_SLAB"))
+    # Measured on MI355X: for 84x84 frames the pinned-cat + SDMA path wins
+    # (~12%) over the GPU-side slab gather, but for full-res frames the
+    # per-batch pinned cat explodes (26+ ms/batch at 3x210x160) and the
+    # slot-id slab path wins by an order of magnitude. Threshold on frame
+    # bytes; TBAMD_OBS_SLAB=1 / TBAMD_NO_OBS_SLAB=1 force either way.
+    frame_bytes = 1
+    for d in (int(x) for x in args.frame.split("x")):
+        frame_bytes *= d
+    use_obs_slab = use_cpp_inference and not os.environ.get(
+        "TBAMD_NO_OBS_SLAB") and (
+        bool(os.environ.get("TBAMD_OBS_SLAB")) or frame_bytes > 48 * 1024)
     pool = runtime.ActorPool(
         unroll_length=T, learner_queue=learner_queue,
         inference_batcher=inference_batcher,

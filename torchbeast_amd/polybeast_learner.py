@@ -427,6 +427,13 @@ def train(flags):  # noqa: C901
         flags.actor_device.type == "cuda"
         and not getattr(flags, "py_inference", False)
     )
+    _shape, _ = observation_shape(flags)
+    _frame_bytes = 1
+    for _d in _shape:
+        _frame_bytes *= _d
+    use_obs_slab = use_cpp_inference and not os.environ.get(
+        "TBAMD_NO_OBS_SLAB") and (
+        bool(os.environ.get("TBAMD_OBS_SLAB")) or _frame_bytes > 48 * 1024)
     actor_pool = runtime.ActorPool(
         unroll_length=flags.unroll_length,
         learner_queue=learner_queue,
@@ -434,10 +441,11 @@ def train(flags):  # noqa: C901
         env_server_addresses=addresses,
         initial_agent_state=initial_agent_state,
         seed_base=rank * flags.num_actors,
-        # Opt-in (TBAMD_OBS_SLAB): actors publish observations into a
-        # pinned slab and requests carry only slot ids (GPU-side gather).
-        # Default off: measured slower than pinned-cat + SDMA on MI355X.
-        use_obs_slab=use_cpp_inference and bool(os.environ.get("TBAMD_OBS_SLAB")),
+        # Actors publish observations into a pinned slab and requests carry
+        # only slot ids (GPU-side gather). Default: on for large frames
+        # (full-res per-batch pinned cat measured 26+ ms), off for 84x84
+        # (pinned-cat + SDMA measured ~12% faster there).
+        use_obs_slab=use_obs_slab,
         rollout_budget_mb=flags.rollout_buffer_budget_mb,
     )
 
@@ -481,7 +489,7 @@ def train(flags):  # noqa: C901
 
     actorpool_thread.start()
     if inference_runner is not None:
-        if os.environ.get("TBAMD_OBS_SLAB"):
+        if use_obs_slab:
             slab = actor_pool.obs_slab()  # blocks until the first env obs
             if slab:
                 inference_runner.set_obs_slab(*slab)
