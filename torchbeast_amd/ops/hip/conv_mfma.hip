@@ -612,13 +612,13 @@ struct F1 {  // conv1 8x8 s4, 3->32: 210x160 -> 51x39
 };
 struct F2 {  // conv2 4x4 s2, 32->64: 51x39 -> 24x18
   static constexpr int CI = 32, IH = 51, IW = 39, KH = 4, KW = 4, ST = 2;
-  static constexpr int CO = 64, OH = 24, OW = 18, SB = 1, OYT = 12;
+  static constexpr int CO = 64, OH = 24, OW = 18, SB = 1, OYT = 6;
   static constexpr size_t LDS =
       (size_t)((OYT - 1) * ST + KH) * IW * CI * 2;
 };
 struct F3 {  // conv3 3x3 s1, 64->64: 24x18 -> 22x16 (flat 22528)
   static constexpr int CI = 64, IH = 24, IW = 18, KH = 3, KW = 3, ST = 1;
-  static constexpr int CO = 64, OH = 22, OW = 16, SB = 1, OYT = 22;
+  static constexpr int CO = 64, OH = 22, OW = 16, SB = 1, OYT = 6;
   static constexpr size_t LDS =
       (size_t)((OYT - 1) * ST + KH) * IW * CI * 2;
 };

@@ -79,6 +79,14 @@ inline torch::Tensor cat_pinned(const std::vector<torch::Tensor>& tensors,
   int64_t total = 0;
   for (const auto& t : tensors) total += t.size(dim);
   shape[dim] = total;
+  int64_t bytes = tensors[0].element_size();
+  for (auto d : shape) bytes *= d;
+  if (bytes < 65536) {
+    // Small outputs (slot ids, scalars): ragged pinned allocations would
+    // hit hipHostMalloc (a device-wide sync) for every new size; a plain
+    // cat + pageable H2D is far cheaper at this size.
+    return torch::cat(tensors, dim);
+  }
   torch::Tensor out = torch::empty(
       shape, tensors[0].options().pinned_memory(true));
   torch::cat_out(out, tensors, dim);

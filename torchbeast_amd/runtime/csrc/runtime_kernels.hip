@@ -147,8 +147,13 @@ std::vector<torch::Tensor> gather_obs(torch::Tensor slab_frames,
   const int b = ids_cpu.numel();
   auto dev = torch::Device(torch::kCUDA, at::cuda::current_device());
   auto opts = torch::TensorOptions().device(dev);
-  auto ids = ids_cpu.to(torch::kInt32).reshape({-1}).pin_memory().to(
-      dev, /*non_blocking=*/true);
+  // Pad the ids staging to the quantized batch size so the pinned-alloc
+  // cache sees a handful of sizes (a fresh hipHostMalloc device-syncs).
+  auto ids_pin = torch::zeros({bp}, torch::TensorOptions()
+                                        .dtype(torch::kInt32)
+                                        .pinned_memory(true));
+  ids_pin.narrow(0, 0, b).copy_(ids_cpu.to(torch::kInt32).reshape({-1}));
+  auto ids = ids_pin.to(dev, /*non_blocking=*/true);
   int64_t fsz = 1;
   for (auto d : frame_shape) fsz *= d;
   std::vector<int64_t> oshape = {bp};
@@ -181,10 +186,16 @@ std::vector<torch::Tensor> fused_heads_sample(torch::Tensor x,
   TORCH_CHECK(A <= 64, "heads kernel supports up to 64 actions");
   TORCH_CHECK(policy_w.size(1) == D + 1 && base_w.size(1) == D + 1,
               "head weights must take [x, reward]");
+  // Allocate pinned outputs at the quantized batch size (cache-friendly;
+  // ragged sizes would hipHostMalloc + device-sync per serve), then hand
+  // back narrowed views.
+  const int64_t bq = (b + 63) / 64 * 64;
   auto hopts = torch::TensorOptions().pinned_memory(true);
-  auto action = torch::empty({b}, hopts.dtype(torch::kInt64));
-  auto logits = torch::empty({b, A}, hopts.dtype(torch::kFloat32));
-  auto baseline = torch::empty({b}, hopts.dtype(torch::kFloat32));
+  auto action = torch::empty({bq}, hopts.dtype(torch::kInt64)).narrow(0, 0, b);
+  auto logits =
+      torch::empty({bq, A}, hopts.dtype(torch::kFloat32)).narrow(0, 0, b);
+  auto baseline =
+      torch::empty({bq}, hopts.dtype(torch::kFloat32)).narrow(0, 0, b);
   const size_t lds = (size_t)(D + 1) * sizeof(float);
   hipLaunchKernelGGL(heads_sample_kernel, dim3((uint32_t)b), dim3(256), lds,
                      at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
