@@ -196,7 +196,24 @@ def main():
         use_obs_slab=use_obs_slab,
         rollout_budget_mb=args.rollout_budget_mb,
     )
-    pool_thread = threading.Thread(target=pool.run, daemon=True)
+    def run_pool():
+        try:
+            pool.run()
+        except Exception:
+            import traceback
+            traceback.print_exc()
+            # Make actor death loud: unblock the learner loop so the bench
+            # fails fast instead of hanging on an empty queue.
+            try:
+                inference_batcher.close()
+            except Exception:
+                pass
+            try:
+                learner_queue.close()
+            except Exception:
+                pass
+
+    pool_thread = threading.Thread(target=run_pool, daemon=True)
     pool_thread.start()
 
     inference_runner = None

@@ -449,12 +449,25 @@ def train(flags):  # noqa: C901
         rollout_budget_mb=flags.rollout_buffer_budget_mb,
     )
 
+    pool_failure = []
+
     def run_pool():
         try:
             actor_pool.run()
-        except Exception:
+        except Exception as e:
+            # Actor death must be LOUD: without this, a dropped first batch
+            # (e.g. a serve error) kills every actor thread silently and
+            # train() starves forever on an empty learner queue.
             logging.exception("Exception in actor pool")
-            raise
+            pool_failure.append(e)
+            try:
+                inference_batcher.close()
+            except Exception:
+                pass
+            try:
+                learner_queue.close()
+            except Exception:
+                pass
 
     actorpool_thread = threading.Thread(target=run_pool, name="actorpool")
 
@@ -520,7 +533,7 @@ def train(flags):  # noqa: C901
         last_checkpoint_time = timer()
         while update_counter["done"] < num_updates and any(
             t.is_alive() for t in learner_threads
-        ):
+        ) and not pool_failure:
             start_time = timer()
             start_step = stats.get("step", 0)
             time.sleep(5)
@@ -559,6 +572,8 @@ def train(flags):  # noqa: C901
         if plogger is not None:
             plogger.close()
 
+    if pool_failure:
+        raise RuntimeError("actor pool failed during training") from pool_failure[0]
     logging.info("Rank %d done after %d updates.", rank, update_counter["done"])
     return model
 

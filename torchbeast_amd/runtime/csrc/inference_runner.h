@@ -127,7 +127,15 @@ class InferenceRunner {
     }
   }
 
+  void trace(const char* what) {
+    if (std::getenv("TBAMD_SERVE_TRACE")) {
+      fprintf(stderr, "[serve %p] %s\n", (void*)this, what);
+      fflush(stderr);
+    }
+  }
+
   void serve(DynamicBatcher::Batch& batch, at::cuda::CUDAStream& stream) {
+    trace("begin");
     // Stage timers (enabled by TBAMD_SERVE_TIMINGS): cat / fwd / d2h+sync.
     const bool timing = serve_timing_;
     auto now_us = []() {
@@ -137,6 +145,7 @@ class InferenceRunner {
     };
     int64_t t0 = timing ? now_us() : 0;
     TensorNest inputs = batch.get_inputs();
+    trace("cat-done");
     int64_t t_cat = timing ? now_us() : 0;
     // Serialize the FIRST forward of each quantized batch size: concurrent
     // MIOpen solution-finds for a brand-new conv shape across streams have
@@ -250,8 +259,10 @@ class InferenceRunner {
         w2p = w2p_;
         w3p = w3p_;
       }
+      trace("trunk");
       x = tbamd::conv_trunk_fwd(frames_p, w1p, weights_[1], w2p, weights_[3],
                                 w3p, weights_[5], /*want_stash=*/false)[0];
+      trace("trunk-done");
     } else if (bp <= 384 && tbamd::atari_trunk_supported(C, H, W)) {
       // Hand-written fused CDNA4 conv trunk: one kernel for the u8
       // normalize + 3 convs (non-84x84 geometries).
@@ -274,11 +285,14 @@ class InferenceRunner {
       // buffers: replaces cat/2x linear/rand/log/argmax/3x D2H.
       const int64_t seed =
           greedy_ ? 0 : (int64_t)(seed_ctr_.fetch_add(1) * 0x9E3779B97F4A7C15ull);
+      trace("heads");
       auto hs = fused_heads_sample(
           x, rew, weights_[head_base_], weights_[head_base_ + 1],
           weights_[head_base_ + 2], weights_[head_base_ + 3], b, greedy_,
           seed);
+      trace("heads-sync");
       stream.synchronize();
+      trace("heads-done");
       int64_t t_fwd2 = timing ? now_us() : 0;
       TensorNest::vector_t agent_out{
           TensorNest(hs[0].reshape({1, b})),
