@@ -14,20 +14,22 @@ if not torch.cuda.is_available():  # pragma: no cover
     pytest.skip("requires ROCm GPU", allow_module_level=True)
 
 
-def test_polybeast_train_short_gpu(tmp_path):
-    from torchbeast_amd import polybeast_learner
+def _repo_root():
+    return os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
-    flags = polybeast_learner.parser.parse_args([])
-    flags.env = "synthetic:4x84x84:6"
-    flags.savedir = str(tmp_path)
-    flags.xpid = "gpue2e"
-    flags.num_actors = 16
-    flags.batch_size = 8
-    flags.unroll_length = 20
-    flags.total_steps = 8 * 20 * 6
-    flags.num_learner_threads = 1
-    flags.num_inference_threads = 1
-    polybeast_learner.train(flags)
+
+def test_polybeast_train_short_gpu(tmp_path):
+    # Subprocess: a fresh HIP context per training run (the long pytest
+    # process accumulates CUDA state that makes in-process training runs
+    # flaky on shared boxes).
+    subprocess.check_call(
+        [sys.executable, "-m", "torchbeast_amd.polybeast_learner",
+         "--env", "synthetic:4x84x84:6", "--savedir", str(tmp_path),
+         "--xpid", "gpue2e", "--num_actors", "16", "--batch_size", "8",
+         "--unroll_length", "20", "--total_steps", str(8 * 20 * 6),
+         "--num_learner_threads", "1", "--num_inference_threads", "1"],
+        cwd=_repo_root(), timeout=300,
+    )
     assert os.path.exists(os.path.join(str(tmp_path), "gpue2e", "model.tar"))
 
 
@@ -54,18 +56,15 @@ def test_graft_smoke():
 
 def test_monobeast_train_short_gpu(tmp_path):
     """Monobeast on GPU: CPU actor processes + shared-memory buffers feed a
-    GPU learner running the fused HIP ops."""
-    from torchbeast_amd import monobeast
-
-    flags = monobeast.parser.parse_args([])
-    flags.env = "synthetic:4x84x84:6"
-    flags.savedir = str(tmp_path)
-    flags.xpid = "monogpu"
-    flags.num_actors = 2
-    flags.num_buffers = 6
-    flags.batch_size = 2
-    flags.unroll_length = 16
-    flags.total_steps = 16 * 2 * 5
-    flags.num_learner_threads = 1
-    monobeast.train(flags)
+    GPU learner running the fused HIP ops. Subprocess so the fork-based
+    actor spawn happens in a fresh process (forking the long pytest
+    process after many HIP tests is flaky)."""
+    subprocess.check_call(
+        [sys.executable, "-m", "torchbeast_amd.monobeast",
+         "--env", "synthetic:4x84x84:6", "--savedir", str(tmp_path),
+         "--xpid", "monogpu", "--num_actors", "2", "--num_buffers", "6",
+         "--batch_size", "2", "--unroll_length", "16",
+         "--total_steps", str(16 * 2 * 5), "--num_learner_threads", "1"],
+        cwd=_repo_root(), timeout=300,
+    )
     assert os.path.exists(os.path.join(str(tmp_path), "monogpu", "model.tar"))
