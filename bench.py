@@ -59,13 +59,16 @@ def parse_args():
     p.add_argument("--inference_timeout_ms", type=int, default=5)
     p.add_argument("--episode_length", type=int, default=1000)
     p.add_argument("--rollout_budget_mb", type=int, default=1024)
+    p.add_argument("--hipgraph", action="store_true",
+                   help="Capture the learner step in a hipGraph (1-GPU).")
     p.add_argument("--py_inference", action="store_true",
                    help="Python inference threads instead of the C++ engine.")
     return p.parse_args()
 
 
 def learner_step(flags, batch_tensors, model, optimizer, scheduler, reducer,
-                 actor_flat, flat_param, inference_runner=None):
+                 actor_flat, flat_param, inference_runner=None,
+                 device_ops_only=False):
     env_outputs = pbl.EnvOutput._make(batch_tensors[:5])
     actor_outputs = pbl.AgentOutput._make(batch_tensors[5:8])
     initial_agent_state = batch_tensors[8:]
@@ -108,9 +111,13 @@ def learner_step(flags, batch_tensors, model, optimizer, scheduler, reducer,
     total_loss.backward()
     reducer.reduce()
     optimizer.step()
-    scheduler.step()
     with torch.no_grad():
         actor_flat.copy_(flat_param)
+    if device_ops_only:
+        # hipGraph capture: host-side updates (LR schedule, runner weight
+        # notification) happen outside the captured region, per replay.
+        return total_loss
+    scheduler.step()
     if inference_runner is not None:
         inference_runner.mark_weights_dirty()
     return total_loss
@@ -240,10 +247,50 @@ def main():
 
     queue_iter = iter(learner_queue)
 
+    graph = None
+    static_batch = None
+    use_graph = args.hipgraph and use_cuda and world_size == 1
+
     # Warmup (fills the pipeline, compiles/caches kernels).
     for _ in range(args.warmup):
         learner_step(flags, get_batch(queue_iter), model, optimizer, scheduler,
                      reducer, actor_flat, flat_param, inference_runner)
+
+    if use_graph:
+        # Capture one whole learner step (fwd + V-trace + loss + backward +
+        # fused RMSProp + behavior sync) into a hipGraph; per-iteration the
+        # batch is copied into the static input slots and the graph
+        # replayed. LR decay flows through a device-resident lr scalar.
+        optimizer.enable_device_lr()
+        static_batch = tuple(t.clone() for t in get_batch(queue_iter))
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                learner_step(flags, static_batch, model, optimizer, scheduler,
+                             reducer, actor_flat, flat_param, None,
+                             device_ops_only=True)
+        torch.cuda.current_stream().wait_stream(side)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            learner_step(flags, static_batch, model, optimizer, scheduler,
+                         reducer, actor_flat, flat_param, None,
+                         device_ops_only=True)
+
+    def run_step():
+        if graph is None:
+            learner_step(flags, get_batch(queue_iter), model, optimizer,
+                         scheduler, reducer, actor_flat, flat_param,
+                         inference_runner)
+            return
+        batch = get_batch(queue_iter)
+        for dst, src in zip(static_batch, batch):
+            dst.copy_(src, non_blocking=True)
+        optimizer.push_lr()
+        graph.replay()
+        scheduler.step()
+        if inference_runner is not None:
+            inference_runner.mark_weights_dirty()
 
     timings = None
     if os.environ.get("TBAMD_BENCH_TIMINGS"):
@@ -259,9 +306,12 @@ def main():
 
     for _ in range(args.steps):
         if timings is None:
-            learner_step(flags, get_batch(queue_iter), model, optimizer,
-                         scheduler, reducer, actor_flat, flat_param,
-                         inference_runner)
+            run_step()
+        elif graph is not None:
+            timings.reset()
+            run_step()
+            torch.cuda.synchronize()
+            timings.time("graph_step")
         else:
             timings.reset()
             batch = get_batch(queue_iter)

@@ -231,7 +231,8 @@ def lstm_unroll(core, x, notdone, state):
 # ---------------------------------------------------------------------------
 
 
-def rmsprop_step(param, grad, square_avg, lr, alpha, eps, clip_norm=None):
+def rmsprop_step(param, grad, square_avg, lr, alpha, eps, clip_norm=None,
+                 lr_tensor=None):
     """One RMSProp step over FLAT tensors, fused with global-norm clipping.
 
     All of param/grad/square_avg are 1-D views over the whole model (see
@@ -242,8 +243,11 @@ def rmsprop_step(param, grad, square_avg, lr, alpha, eps, clip_norm=None):
         return ext.rmsprop_step(
             param, grad, square_avg, float(lr), float(alpha), float(eps),
             float(clip_norm) if clip_norm is not None else -1.0,
+            lr_tensor,
         )
 
+    if lr_tensor is not None:
+        lr = float(lr_tensor.item())
     total_norm = grad.norm(2)
     if clip_norm is not None:
         coef = clip_norm / (total_norm + 1e-6)

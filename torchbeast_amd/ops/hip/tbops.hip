@@ -273,13 +273,17 @@ __global__ void grad_norm_sq_kernel(const float* __restrict__ grad, int64_t n,
 
 __global__ void rmsprop_update_kernel(
     float* __restrict__ param, const float* __restrict__ grad,
-    float* __restrict__ square_avg, int64_t n, float lr, float alpha,
-    float eps, float clip_norm, const float* __restrict__ norm_sq,
-    float* __restrict__ norm_out) {
+    float* __restrict__ square_avg, int64_t n, float lr,
+    const float* __restrict__ lr_ptr,  // overrides lr when non-null
+    float alpha, float eps, float clip_norm,
+    const float* __restrict__ norm_sq, float* __restrict__ norm_out) {
   const float norm = sqrtf(*norm_sq);
   float coef = 1.f;
   if (clip_norm > 0.f) coef = fminf(1.f, clip_norm / (norm + 1e-6f));
   if (blockIdx.x == 0 && threadIdx.x == 0) *norm_out = norm;
+  // hipGraph capture freezes by-value arguments; when the learning rate
+  // must keep decaying across graph replays it is read from device memory.
+  if (lr_ptr != nullptr) lr = *lr_ptr;
 
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
@@ -293,11 +297,14 @@ __global__ void rmsprop_update_kernel(
 
 torch::Tensor rmsprop_step(torch::Tensor param, torch::Tensor grad,
                            torch::Tensor square_avg, double lr, double alpha,
-                           double eps, double clip_norm) {
+                           double eps, double clip_norm,
+                           std::optional<torch::Tensor> lr_tensor = {}) {
   TORCH_CHECK(param.is_cuda() && param.dim() == 1, "rmsprop: flat GPU tensor");
   const int64_t n = param.numel();
   auto norm_sq = torch::zeros({1}, param.options());
   auto norm_out = torch::empty({}, param.options());
+  const float* lr_p =
+      lr_tensor ? lr_tensor->data_ptr<float>() : nullptr;
 
   const int threads = 256;
   const int blocks = std::min<int64_t>(2048, ceil_div(n, threads));
@@ -307,8 +314,8 @@ torch::Tensor rmsprop_step(torch::Tensor param, torch::Tensor grad,
                      norm_sq.data_ptr<float>());
   hipLaunchKernelGGL(rmsprop_update_kernel, dim3(blocks), dim3(threads), 0,
                      stream, param.data_ptr<float>(), grad.data_ptr<float>(),
-                     square_avg.data_ptr<float>(), n, (float)lr, (float)alpha,
-                     (float)eps, (float)clip_norm,
+                     square_avg.data_ptr<float>(), n, (float)lr, lr_p,
+                     (float)alpha, (float)eps, (float)clip_norm,
                      norm_sq.data_ptr<float>(), norm_out.data_ptr<float>());
   return norm_out;
 }
@@ -536,7 +543,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_gemm_v2", &tbamd::mfma_gemm_v2);
   m.def("vtrace_from_logits", &vtrace_from_logits);
   m.def("fused_impala_loss_fwd", &fused_impala_loss_fwd);
-  m.def("rmsprop_step", &rmsprop_step);
+  m.def("rmsprop_step", &rmsprop_step, py::arg("param"), py::arg("grad"),
+        py::arg("square_avg"), py::arg("lr"), py::arg("alpha"),
+        py::arg("eps"), py::arg("clip_norm"),
+        py::arg("lr_tensor") = std::nullopt);
   m.def("policy_sample", &policy_sample);
   m.def("lstm_unroll_fwd", &lstm_unroll_fwd);
   m.def("lstm_unroll_bwd", &lstm_unroll_bwd);

@@ -77,6 +77,7 @@ class FusedRMSProp:
         self.lr_factor = 1.0
         self.steps = 0
         self.last_grad_norm = None
+        self.lr_dev = None  # device-resident lr (hipGraph capture mode)
 
     @property
     def lr(self):
@@ -84,6 +85,18 @@ class FusedRMSProp:
 
     def zero_grad(self):
         self.grad.zero_()
+
+    def enable_device_lr(self):
+        """hipGraph mode: the update kernel reads lr from device memory so
+        the linear decay keeps applying across graph replays (by-value
+        kernel args are frozen at capture)."""
+        self.lr_dev = torch.zeros(1, device=self.param.device,
+                                  dtype=torch.float32)
+        self.push_lr()
+
+    def push_lr(self):
+        if self.lr_dev is not None:
+            self.lr_dev.fill_(self.lr)
 
     def step(self):
         self.last_grad_norm = tbops.rmsprop_step(
@@ -94,6 +107,7 @@ class FusedRMSProp:
             self.alpha,
             self.eps,
             self.clip_norm,
+            self.lr_dev,
         )
         self.steps += 1
 
