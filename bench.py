@@ -272,7 +272,9 @@ def main():
                              device_ops_only=True)
         torch.cuda.current_stream().wait_stream(side)
         graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
+        # thread_local capture: the inference engine and queue threads keep
+        # synchronizing their own streams while we capture ours.
+        with torch.cuda.graph(graph, capture_error_mode="thread_local"):
             learner_step(flags, static_batch, model, optimizer, scheduler,
                          reducer, actor_flat, flat_param, None,
                          device_ops_only=True)
