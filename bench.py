@@ -250,7 +250,11 @@ def main():
 
     graph = None
     static_batch = None
-    use_graph = args.hipgraph and use_cuda and world_size == 1
+    # Default hipGraph only for the shallow non-LSTM headline config: the
+    # LSTM's cooperative persistent kernels capture but replay ~3x slower,
+    # and the deep model sees no benefit (measured).
+    use_graph = (args.hipgraph and use_cuda and world_size == 1
+                 and args.model == "shallow" and not args.use_lstm)
 
     # Warmup (fills the pipeline, compiles/caches kernels).
     for _ in range(args.warmup):
