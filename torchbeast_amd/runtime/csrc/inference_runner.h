@@ -231,8 +231,8 @@ class InferenceRunner {
         }
       }
       x = at::relu(x).reshape({bp, -1});
-    } else if ((C == 4 && H == 84 && W == 84) ||
-               (C == 3 && H == 210 && W == 160)) {
+    } else if (!aten_only_ && ((C == 4 && H == 84 && W == 84) ||
+                               (C == 3 && H == 210 && W == 160))) {
       // MFMA implicit-GEMM trunk (bf16 operands, fp32 accumulate) — flat
       // per-batch cost at every dynamic batch size, unlike the per-sample
       // fused kernel whose grid is the batch (underfills the 256 CUs below
@@ -263,7 +263,8 @@ class InferenceRunner {
       x = tbamd::conv_trunk_fwd(frames_p, w1p, weights_[1], w2p, weights_[3],
                                 w3p, weights_[5], /*want_stash=*/false)[0];
       trace("trunk-done");
-    } else if (bp <= 384 && tbamd::atari_trunk_supported(C, H, W)) {
+    } else if (!aten_only_ && bp <= 384 &&
+               tbamd::atari_trunk_supported(C, H, W)) {
       // Hand-written fused CDNA4 conv trunk: one kernel for the u8
       // normalize + 3 convs (non-84x84 geometries).
       x = tbamd::atari_trunk_fwd(frames_p, weights_[0], weights_[1],
@@ -279,7 +280,8 @@ class InferenceRunner {
     x = at::linear(x, weights_[head_base_ - 2], weights_[head_base_ - 1])
             .relu_();
 
-    if (num_lstm_layers_ == 0 && weights_[head_base_].size(0) <= 64 &&
+    if (!aten_only_ && num_lstm_layers_ == 0 &&
+        weights_[head_base_].size(0) <= 64 &&
         weights_[head_base_].size(1) == x.size(1) + 1) {
       // Fused heads + Gumbel sample, written straight into pinned host
       // buffers: replaces cat/2x linear/rand/log/argmax/3x D2H.
@@ -436,6 +438,9 @@ class InferenceRunner {
   int64_t packed_version_ = -1;
   std::atomic<int64_t> weights_version_{0};
   std::atomic<uint64_t> seed_ctr_{1};
+  // Benchmarking: serve everything through library ops (the strategy
+  // comparison in scripts/inference_speed_profiling.py).
+  const bool aten_only_ = std::getenv("TBAMD_RUNNER_ATEN") != nullptr;
   const bool serve_timing_ = std::getenv("TBAMD_SERVE_TIMINGS") != nullptr;
   std::atomic<int64_t> t_cat_us_{0}, t_fwd_us_{0}, t_d2h_us_{0};
   std::atomic<int64_t> timed_batches_{0};
