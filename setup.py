@@ -72,6 +72,7 @@ setup(
         "torchbeast_amd.runtime",
         "torchbeast_amd.nest",
         "torchbeast_amd.parallel",
+        "nest",
     ],
     ext_modules=ext_modules,
     cmdclass={"build_ext": BuildExtension},
