@@ -70,6 +70,9 @@ def make_parser():
     parser.add_argument("--inference_max_batch_size", default=512, type=int)
     parser.add_argument("--inference_timeout_ms", default=10, type=int)
     parser.add_argument("--write_profiler_trace", action="store_true")
+    parser.add_argument("--use_hipgraph", action="store_true",
+                        help="Capture the learner step in a hipGraph "
+                             "(1 GPU, 1 learner thread, shallow non-LSTM).")
     parser.add_argument("--py_inference", action="store_true",
                         help="Serve inference from Python threads instead of "
                              "the C++ engine (always the case for --model "
@@ -217,6 +220,80 @@ def learn(
     """Learner loop: exactly num_updates optimizer steps (shared across
     threads via update_counter), identical on every DP rank so collectives
     stay aligned."""
+
+    def device_step(env_outputs, actor_outputs, initial_agent_state):
+        """Everything between batch-on-GPU and host-side bookkeeping: model
+        forward, V-trace, losses, backward, all-reduce, fused optimizer,
+        behavior-model sync. Pure device work so it is hipGraph-capturable."""
+        learner_outputs, _ = model(
+            dict(
+                frame=env_outputs.frame,
+                reward=env_outputs.rewards,
+                done=env_outputs.done,
+            ),
+            initial_agent_state,
+        )
+        learner_outputs = AgentOutput._make(_as_agent_output(learner_outputs))
+
+        bootstrap_value = learner_outputs.baseline[-1]
+
+        # Shift: env_outputs[t+1] is the consequence of actions[t].
+        env_outputs = EnvOutput._make(t[1:] for t in env_outputs)
+        actor_outputs = AgentOutput._make(t[1:] for t in actor_outputs)
+        learner_outputs = AgentOutput._make(t[:-1] for t in learner_outputs)
+
+        if flags.reward_clipping == "abs_one":
+            clipped_rewards = torch.clamp(env_outputs.rewards, -1, 1)
+        else:
+            clipped_rewards = env_outputs.rewards
+
+        discounts = (~env_outputs.done).float() * flags.discounting
+
+        vtrace_returns = vtrace.from_logits(
+            behavior_policy_logits=actor_outputs.policy_logits,
+            target_policy_logits=learner_outputs.policy_logits,
+            actions=actor_outputs.action,
+            discounts=discounts,
+            rewards=clipped_rewards,
+            values=learner_outputs.baseline,
+            bootstrap_value=bootstrap_value,
+        )
+
+        pg_loss, baseline_loss, entropy_loss = tbops.fused_impala_loss(
+            learner_outputs.policy_logits,
+            learner_outputs.baseline,
+            actor_outputs.action,
+            vtrace_returns.pg_advantages,
+            vtrace_returns.vs,
+        )
+        total_loss = (
+            pg_loss
+            + flags.baseline_cost * baseline_loss
+            + flags.entropy_cost * entropy_loss
+        )
+
+        optimizer.zero_grad()
+        total_loss.backward()
+        reducer.reduce()
+        optimizer.step()
+
+        # Behavior-model sync: one flat copy (D2D on GPU).
+        with torch.no_grad():
+            actor_flat.copy_(flat_param)
+        return total_loss, pg_loss, baseline_loss, entropy_loss
+
+    use_graph = (
+        getattr(flags, "use_hipgraph", False)
+        and flags.learner_device.type == "cuda"
+        and reducer.world_size == 1
+        and flags.num_learner_threads == 1
+        and flags.model == "shallow"
+        and not flags.use_lstm
+    )
+    graph = None
+    static_in = None
+    static_out = None
+
     for tensors in learner_queue:
         tensors = tuple(
             t.to(flags.learner_device, non_blocking=True)
@@ -233,62 +310,52 @@ def learn(
                     return
                 update_counter["done"] += 1
 
-            learner_outputs, _ = model(
-                dict(
-                    frame=env_outputs.frame,
-                    reward=env_outputs.rewards,
-                    done=env_outputs.done,
-                ),
-                initial_agent_state,
-            )
-            learner_outputs = AgentOutput._make(_as_agent_output(learner_outputs))
+            if use_graph and graph is None:
+                # First batch on this thread: capture the device step.
+                optimizer.enable_device_lr()
+                static_in = (
+                    EnvOutput._make(t.clone() for t in env_outputs),
+                    AgentOutput._make(t.clone() for t in actor_outputs),
+                    tuple(t.clone() for t in initial_agent_state),
+                )
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(3):
+                        device_step(*static_in)
+                torch.cuda.current_stream().wait_stream(side)
+                try:
+                    graph = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(
+                        graph, capture_error_mode="thread_local"
+                    ):
+                        static_out = device_step(*static_in)
+                except Exception:
+                    logging.exception(
+                        "hipGraph capture failed; continuing eager")
+                    use_graph = False
+                    graph = None
+                    torch.cuda.synchronize()
 
-            bootstrap_value = learner_outputs.baseline[-1]
-
-            # Shift: env_outputs[t+1] is the consequence of actions[t].
-            env_outputs = EnvOutput._make(t[1:] for t in env_outputs)
-            actor_outputs = AgentOutput._make(t[1:] for t in actor_outputs)
-            learner_outputs = AgentOutput._make(t[:-1] for t in learner_outputs)
-
-            if flags.reward_clipping == "abs_one":
-                clipped_rewards = torch.clamp(env_outputs.rewards, -1, 1)
+            if graph is not None:
+                for dst, src in zip(
+                    list(static_in[0]) + list(static_in[1])
+                    + list(static_in[2]),
+                    list(env_outputs) + list(actor_outputs)
+                    + list(initial_agent_state),
+                ):
+                    dst.copy_(src, non_blocking=True)
+                optimizer.push_lr()
+                graph.replay()
+                total_loss, pg_loss, baseline_loss, entropy_loss = static_out
+                env_outputs = EnvOutput._make(
+                    t[1:] for t in static_in[0])
             else:
-                clipped_rewards = env_outputs.rewards
-
-            discounts = (~env_outputs.done).float() * flags.discounting
-
-            vtrace_returns = vtrace.from_logits(
-                behavior_policy_logits=actor_outputs.policy_logits,
-                target_policy_logits=learner_outputs.policy_logits,
-                actions=actor_outputs.action,
-                discounts=discounts,
-                rewards=clipped_rewards,
-                values=learner_outputs.baseline,
-                bootstrap_value=bootstrap_value,
-            )
-
-            pg_loss, baseline_loss, entropy_loss = tbops.fused_impala_loss(
-                learner_outputs.policy_logits,
-                learner_outputs.baseline,
-                actor_outputs.action,
-                vtrace_returns.pg_advantages,
-                vtrace_returns.vs,
-            )
-            total_loss = (
-                pg_loss
-                + flags.baseline_cost * baseline_loss
-                + flags.entropy_cost * entropy_loss
-            )
-
-            optimizer.zero_grad()
-            total_loss.backward()
-            reducer.reduce()
-            optimizer.step()
+                (total_loss, pg_loss, baseline_loss,
+                 entropy_loss) = device_step(env_outputs, actor_outputs,
+                                             initial_agent_state)
+                env_outputs = EnvOutput._make(t[1:] for t in env_outputs)
             scheduler.step()
-
-            # Behavior-model sync: one flat copy (D2D on GPU).
-            with torch.no_grad():
-                actor_flat.copy_(flat_param)
             if inference_runner is not None:
                 inference_runner.mark_weights_dirty()
 
