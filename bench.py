@@ -310,6 +310,10 @@ def main():
         from torchbeast_amd.core.prof import Timings
 
         timings = Timings()
+        # Steady-state stats: drop the startup transients accumulated
+        # during warmup (first-serve compiles, allocator growth).
+        learner_queue.reset_stats()
+        inference_batcher.reset_stats()
 
     if world_size > 1:
         torch.distributed.barrier()

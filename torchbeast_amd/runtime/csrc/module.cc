@@ -176,6 +176,7 @@ PYBIND11_MODULE(_tbruntime, m) {
            py::call_guard<py::gil_scoped_release>())
       .def("size", &BatchingQueue::size)
       .def("stats", &BatchingQueue::stats)
+      .def("reset_stats", &BatchingQueue::reset_stats)
       .def("close", &BatchingQueue::close,
            py::call_guard<py::gil_scoped_release>())
       .def("is_closed", &BatchingQueue::is_closed)
@@ -216,6 +217,7 @@ PYBIND11_MODULE(_tbruntime, m) {
            py::call_guard<py::gil_scoped_release>())
       .def("size", &DynamicBatcher::size)
       .def("stats", &DynamicBatcher::stats)
+      .def("reset_stats", &DynamicBatcher::reset_stats)
       .def("close", &DynamicBatcher::close,
            py::call_guard<py::gil_scoped_release>())
       .def("is_closed", &DynamicBatcher::is_closed)

@@ -47,6 +47,10 @@ struct StageStats {
     n.fetch_add(1, std::memory_order_relaxed);
     sum_us.fetch_add(us, std::memory_order_relaxed);
   }
+  void reset() {
+    n.store(0, std::memory_order_relaxed);
+    sum_us.store(0, std::memory_order_relaxed);
+  }
 };
 
 class ClosedQueue : public std::runtime_error {
@@ -498,6 +502,12 @@ class BatchingQueue {
     source_pool_ = std::move(pool);
   }
 
+  // Drop accumulated latency stats (steady-state measurement after warmup).
+  void reset_stats() {
+    wait_stats_.reset();
+    cat_stats_.reset();
+  }
+
  private:
   const std::optional<torch::Device> output_device_;
   const int64_t batch_dim_;
@@ -697,6 +707,13 @@ class DynamicBatcher {
       out["avg_service_ms"] = service_stats_.sum_us / 1e3 / service_stats_.n;
     }
     return out;
+  }
+
+  void reset_stats() {
+    roundtrip_stats_.reset();
+    formation_stats_.reset();
+    batch_size_stats_.reset();
+    service_stats_.reset();
   }
 
  private:
