@@ -59,6 +59,10 @@ def parse_args():
     p.add_argument("--inference_timeout_ms", type=int, default=5)
     p.add_argument("--episode_length", type=int, default=1000)
     p.add_argument("--rollout_budget_mb", type=int, default=1024)
+    p.add_argument("--envs_per_thread", type=int, default=8,
+                   help="Env streams per actor thread (event-driven loop; "
+                        "512 one-env threads measured host-scheduling "
+                        "bound).")
     p.add_argument("--hipgraph", action="store_true", default=True,
                    help="Capture the learner step in a hipGraph (1-GPU).")
     p.add_argument("--no_hipgraph", dest="hipgraph", action="store_false")
@@ -203,6 +207,7 @@ def main():
         seed_base=rank * args.actors,  # distinct env streams per rank
         use_obs_slab=use_obs_slab,
         rollout_budget_mb=args.rollout_budget_mb,
+        envs_per_thread=args.envs_per_thread,
     )
     def run_pool():
         try:

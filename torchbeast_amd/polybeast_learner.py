@@ -58,6 +58,10 @@ def make_parser():
     parser.add_argument("--num_learner_threads", default=2, type=int)
     parser.add_argument("--num_inference_threads", default=2, type=int)
     parser.add_argument("--max_learner_queue_size", default=None, type=int)
+    parser.add_argument("--envs_per_thread", default=0, type=int,
+                        help="Env streams per actor thread (0 = auto: 8 for "
+                             "in-process synthetic envs, 1 for socket envs "
+                             "so a slow remote env cannot stall neighbors).")
     parser.add_argument("--rollout_buffer_budget_mb", default=1024, type=int,
                         help="Pinned rollout ring budget (MB); actors block "
                              "when it is exhausted (0 = unbounded ad-hoc "
@@ -514,6 +518,10 @@ def train(flags):  # noqa: C901
         # (pinned-cat + SDMA measured ~12% faster there).
         use_obs_slab=use_obs_slab,
         rollout_budget_mb=flags.rollout_buffer_budget_mb,
+        envs_per_thread=(
+            flags.envs_per_thread if flags.envs_per_thread > 0
+            else (8 if tbflags.parse_synthetic_env_spec(flags.env) is not None
+                  else 1)),
     )
 
     pool_failure = []

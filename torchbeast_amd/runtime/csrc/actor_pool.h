@@ -29,14 +29,16 @@ class ActorPool {
             std::shared_ptr<DynamicBatcher> inference_batcher,
             std::vector<std::string> env_server_addresses,
             TensorNest initial_agent_state, int64_t seed_base = 0,
-            bool use_obs_slab = false, int64_t rollout_budget_mb = 0)
+            bool use_obs_slab = false, int64_t rollout_budget_mb = 0,
+            int64_t envs_per_thread = 1)
       : unroll_length_(unroll_length),
         learner_queue_(std::move(learner_queue)),
         inference_batcher_(std::move(inference_batcher)),
         addresses_(std::move(env_server_addresses)),
         initial_agent_state_(std::move(initial_agent_state)),
         seed_base_(seed_base),
-        use_obs_slab_(use_obs_slab) {
+        use_obs_slab_(use_obs_slab),
+        envs_per_thread_(std::max<int64_t>(1, envs_per_thread)) {
     if (unroll_length_ < 1) {
       throw std::invalid_argument("unroll_length must be >= 1");
     }
@@ -74,6 +76,36 @@ class ActorPool {
   // Rethrows the first actor failure.
   void run() {
     std::vector<std::future<void>> futures;
+    const size_t k = (size_t)envs_per_thread_;
+    if (k > 1) {
+      // Event-driven mode: each thread drives k env streams with
+      // overlapped inference requests. 512 actor threads were measured
+      // host-scheduling-bound (serve latency scaled with thread count,
+      // profiles/PROFILE_r2.md); k envs per thread keeps the same env
+      // parallelism with 1/k threads. Best for in-process synthetic envs
+      // whose step() is microseconds; socket envs keep k == 1 so a slow
+      // remote env cannot stall its neighbors.
+      const size_t nthreads = (addresses_.size() + k - 1) / k;
+      futures.reserve(nthreads);
+      for (size_t t = 0; t < nthreads; ++t) {
+        const size_t lo = t * k;
+        const size_t hi = std::min(addresses_.size(), lo + k);
+        futures.push_back(std::async(std::launch::async, [this, lo, hi] {
+          try {
+            loop_multi(lo, hi);
+          } catch (...) {
+            {
+              std::lock_guard<std::mutex> lk(slab_mu_);
+              failed_ = true;
+            }
+            slab_cv_.notify_all();
+            throw;
+          }
+        }));
+      }
+      collect(futures);
+      return;
+    }
     futures.reserve(addresses_.size());
     for (size_t i = 0; i < addresses_.size(); ++i) {
       futures.push_back(std::async(std::launch::async, [this, i] {
@@ -90,17 +122,7 @@ class ActorPool {
         }
       }));
     }
-    std::exception_ptr first_error;
-    for (auto& f : futures) {
-      try {
-        f.get();
-      } catch (const ClosedQueue&) {
-        // Normal shutdown path.
-      } catch (...) {
-        if (!first_error) first_error = std::current_exception();
-      }
-    }
-    if (first_error) std::rethrow_exception(first_error);
+    collect(futures);
   }
 
   uint64_t count() const { return step_count_.load(std::memory_order_relaxed); }
@@ -135,6 +157,161 @@ class ActorPool {
         frame.data_ptr(), fsz);
     slab_rew_.data_ptr<float>()[slot] = f[1].leaf().item<float>();
     slab_done_.data_ptr<uint8_t>()[slot] = f[2].leaf().item<bool>() ? 1 : 0;
+  }
+
+  void collect(std::vector<std::future<void>>& futures) {
+    std::exception_ptr first_error;
+    for (auto& f : futures) {
+      try {
+        f.get();
+      } catch (const ClosedQueue&) {
+        // Normal shutdown path.
+      } catch (...) {
+        if (!first_error) first_error = std::current_exception();
+      }
+    }
+    if (first_error) std::rethrow_exception(first_error);
+  }
+
+  // One thread, several env streams, overlapped inference futures. A
+  // request is outstanding per env; completions are collected with a
+  // short-timeout wait on one pending future plus a zero-timeout scan of
+  // the rest (completions are correlated: batchmates finish together).
+  void loop_multi(size_t lo, size_t hi) {
+    struct EnvSlot {
+      std::unique_ptr<EnvConnection> env;
+      TensorNest env_outputs;
+      TensorNest agent_state;
+      TensorNest rollout_initial_state;
+      TensorNest state_before;
+      TensorNest slot_req;
+      std::vector<TensorNest> rollout;
+      std::future<TensorNest> pending;
+      bool has_pending = false;
+    };
+    const size_t n = hi - lo;
+    std::vector<EnvSlot> envs(n);
+    for (size_t i = 0; i < n; ++i) {
+      EnvSlot& e = envs[i];
+      e.env = make_env_connection(addresses_[lo + i],
+                                  seed_base_ + (int64_t)(lo + i) + 1);
+      e.env_outputs = e.env->initial();
+      e.agent_state = initial_agent_state_;
+      e.rollout_initial_state = initial_agent_state_;
+      e.rollout.reserve(unroll_length_ + 1);
+      if (use_obs_slab_) {
+        e.slot_req = TensorNest(torch::full(
+            {1, 1}, (int64_t)(lo + i),
+            torch::TensorOptions().dtype(torch::kInt32)));
+      }
+    }
+
+    auto issue = [&](size_t i) {
+      EnvSlot& e = envs[i];
+      e.state_before = e.agent_state;
+      TensorNest request;
+      if (use_obs_slab_) {
+        fill_slot((int64_t)(lo + i), e.env_outputs);
+        request = TensorNest(TensorNest::vector_t{e.slot_req, e.agent_state});
+      } else {
+        request =
+            TensorNest(TensorNest::vector_t{e.env_outputs, e.agent_state});
+      }
+      e.pending = inference_batcher_->compute_async(std::move(request));
+      e.has_pending = true;
+    };
+
+    for (size_t i = 0; i < n; ++i) issue(i);
+
+    for (;;) {
+      // Block briefly on the first pending future, then sweep all.
+      size_t first = n;
+      for (size_t i = 0; i < n; ++i) {
+        if (envs[i].has_pending) {
+          first = i;
+          break;
+        }
+      }
+      if (first == n) throw std::logic_error("no pending inference");
+      envs[first].pending.wait_for(std::chrono::microseconds(500));
+
+      for (size_t i = 0; i < n; ++i) {
+        EnvSlot& e = envs[i];
+        if (!e.has_pending ||
+            e.pending.wait_for(std::chrono::seconds(0)) !=
+                std::future_status::ready) {
+          continue;
+        }
+        e.has_pending = false;
+        TensorNest result = e.pending.get();  // rethrows AsyncError
+        if (!result.is_vector() || result.vector().size() != 2) {
+          throw std::runtime_error(
+              "inference must return ((action, ...), agent_state)");
+        }
+        TensorNest agent_outputs = result.vector()[0];
+        e.agent_state = result.vector()[1];
+
+        if (e.rollout.empty()) e.rollout_initial_state = e.state_before;
+        e.rollout.push_back(
+            TensorNest(TensorNest::vector_t{e.env_outputs, agent_outputs}));
+
+        if (static_cast<int64_t>(e.rollout.size()) == unroll_length_ + 1) {
+          flush_rollout(e.rollout, e.rollout_initial_state);
+          TensorNest last = std::move(e.rollout.back());
+          e.rollout.clear();
+          e.rollout.push_back(std::move(last));
+          e.rollout_initial_state = e.state_before;
+        }
+
+        const torch::Tensor& action = agent_outputs.front();
+        e.env_outputs = e.env->step(action);
+        step_count_.fetch_add(1, std::memory_order_relaxed);
+        issue(i);
+      }
+    }
+  }
+
+  // Stack a complete [T+1] rollout into a pinned destination and enqueue.
+  void flush_rollout(const std::vector<TensorNest>& rollout,
+                     const TensorNest& rollout_initial_state) {
+    std::vector<const TensorNest*> steps;
+    steps.reserve(rollout.size());
+    for (const auto& s : rollout) steps.push_back(&s);
+    TensorNest stacked;
+    if (rollout_pool_) {
+      if (rollout_slot_bytes_ == 0) {
+        int64_t total = 0;
+        TensorNest::apply_columns(
+            steps, [&total](const std::vector<torch::Tensor>& column) {
+              int64_t rows = 0;
+              for (const auto& t : column) rows += t.size(0);
+              const int64_t bytes =
+                  rows * column[0].stride(0) * column[0].element_size();
+              total += (bytes + 255) & ~int64_t(255);
+              return column[0];
+            });
+        rollout_slot_bytes_ = total + 4096;
+      }
+      auto slot = rollout_pool_->acquire(
+          rollout_slot_bytes_, [this] { return learner_queue_->is_closed(); });
+      stacked = TensorNest::apply_columns(
+          steps, [&slot](const std::vector<torch::Tensor>& column) {
+            auto shape = column[0].sizes().vec();
+            int64_t rows = 0;
+            for (const auto& t : column) rows += t.size(0);
+            shape[0] = rows;
+            torch::Tensor out = slot.carve(shape, column[0].scalar_type());
+            torch::cat_out(out, column, 0);
+            return out;
+          });
+    } else {
+      stacked = TensorNest::apply_columns(
+          steps, [](const std::vector<torch::Tensor>& column) {
+            return cat_pinned(column, /*dim=*/0);
+          });
+    }
+    learner_queue_->enqueue(TensorNest(
+        TensorNest::vector_t{std::move(stacked), rollout_initial_state}));
   }
 
   void loop(const std::string& address, uint64_t seed) {
@@ -177,50 +354,7 @@ class ActorPool {
           TensorNest(TensorNest::vector_t{env_outputs, agent_outputs}));
 
       if (static_cast<int64_t>(rollout.size()) == unroll_length_ + 1) {
-        std::vector<const TensorNest*> steps;
-        steps.reserve(rollout.size());
-        for (const auto& s : rollout) steps.push_back(&s);
-        // Pinned destination so the learner-side dequeue can DMA it to the
-        // GPU: a slot of the budget-bounded ring when configured, else an
-        // ad-hoc pinned allocation.
-        TensorNest stacked;
-        if (rollout_pool_) {
-          if (rollout_slot_bytes_ == 0) {
-            int64_t total = 0;
-            TensorNest::apply_columns(
-                steps, [&total](const std::vector<torch::Tensor>& column) {
-                  int64_t rows = 0;
-                  for (const auto& t : column) rows += t.size(0);
-                  const int64_t bytes = rows * column[0].stride(0) *
-                                        column[0].element_size();
-                  total += (bytes + 255) & ~int64_t(255);
-                  return column[0];
-                });
-            rollout_slot_bytes_ = total + 4096;
-          }
-          auto slot = rollout_pool_->acquire(
-              rollout_slot_bytes_,
-              [this] { return learner_queue_->is_closed(); });
-          stacked = TensorNest::apply_columns(
-              steps, [&slot](const std::vector<torch::Tensor>& column) {
-                auto shape = column[0].sizes().vec();
-                int64_t rows = 0;
-                for (const auto& t : column) rows += t.size(0);
-                shape[0] = rows;
-                torch::Tensor out =
-                    slot.carve(shape, column[0].scalar_type());
-                torch::cat_out(out, column, 0);
-                return out;
-              });
-        } else {
-          stacked = TensorNest::apply_columns(
-              steps, [](const std::vector<torch::Tensor>& column) {
-                return cat_pinned(column, /*dim=*/0);
-              });
-        }
-        learner_queue_->enqueue(TensorNest(
-            TensorNest::vector_t{std::move(stacked), rollout_initial_state}));
-
+        flush_rollout(rollout, rollout_initial_state);
         // Overlap by one step: the rollout we just sent ends with the step
         // whose pre-inference state is `state_before`.
         TensorNest last = std::move(rollout.back());
@@ -242,6 +376,7 @@ class ActorPool {
   TensorNest initial_agent_state_;
   const int64_t seed_base_;
   const bool use_obs_slab_;
+  const int64_t envs_per_thread_;
   std::mutex slab_mu_;
   std::condition_variable slab_cv_;
   std::atomic<bool> slab_ready_{false};

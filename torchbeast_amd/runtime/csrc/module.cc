@@ -239,12 +239,13 @@ PYBIND11_MODULE(_tbruntime, m) {
   py::class_<ActorPool, std::shared_ptr<ActorPool>>(m, "ActorPool")
       .def(py::init<int64_t, std::shared_ptr<BatchingQueue>,
                     std::shared_ptr<DynamicBatcher>, std::vector<std::string>,
-                    TensorNest, int64_t, bool, int64_t>(),
+                    TensorNest, int64_t, bool, int64_t, int64_t>(),
            py::arg("unroll_length"), py::arg("learner_queue"),
            py::arg("inference_batcher"), py::arg("env_server_addresses"),
            py::arg("initial_agent_state"), py::arg("seed_base") = 0,
            py::arg("use_obs_slab") = false,
-           py::arg("rollout_budget_mb") = 0)
+           py::arg("rollout_budget_mb") = 0,
+           py::arg("envs_per_thread") = 1)
       .def("run", &ActorPool::run, py::call_guard<py::gil_scoped_release>())
       .def("obs_slab", &ActorPool::obs_slab,
            py::call_guard<py::gil_scoped_release>())

@@ -644,7 +644,9 @@ class DynamicBatcher {
         queue_(std::nullopt) {}
 
   // Called by actor threads; blocks until the consumer sets outputs.
-  TensorNest compute(TensorNest inputs) {
+  // Enqueue a request and return the future (the multi-env actor loop
+  // waits on several of these at once).
+  std::future<TensorNest> compute_async(TensorNest inputs) {
     if (inputs.empty()) {
       throw std::invalid_argument("compute() on empty nest");
     }
@@ -662,8 +664,13 @@ class DynamicBatcher {
     });
     auto promise = std::make_shared<std::promise<TensorNest>>();
     std::future<TensorNest> future = promise->get_future();
-    const int64_t t0 = now_us();
     queue_.enqueue(Request{std::move(inputs), batch_size, std::move(promise)});
+    return future;
+  }
+
+  TensorNest compute(TensorNest inputs) {
+    const int64_t t0 = now_us();
+    std::future<TensorNest> future = compute_async(std::move(inputs));
     if (future.wait_for(std::chrono::minutes(10)) ==
         std::future_status::timeout) {
       throw AsyncError("compute() timed out after 10 minutes");
