@@ -46,8 +46,9 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--batch_size", type=int, default=32)
     p.add_argument("--unroll_length", type=int, default=80)
-    p.add_argument("--actors", type=int, default=512,
-                   help="Env streams per GPU (tuned on MI355X; see profiles/).")
+    p.add_argument("--actors", type=int, default=640,
+                   help="Env streams per GPU (tuned on MI355X; see "
+                        "profiles/PROFILE_r2.md).")
     p.add_argument("--model", default="shallow", choices=["shallow", "deep"])
     p.add_argument("--use_lstm", action="store_true")
     p.add_argument("--frame", default="4x84x84",
@@ -59,7 +60,7 @@ def parse_args():
     p.add_argument("--inference_timeout_ms", type=int, default=5)
     p.add_argument("--episode_length", type=int, default=1000)
     p.add_argument("--rollout_budget_mb", type=int, default=1024)
-    p.add_argument("--envs_per_thread", type=int, default=8,
+    p.add_argument("--envs_per_thread", type=int, default=16,
                    help="Env streams per actor thread (event-driven loop; "
                         "512 one-env threads measured host-scheduling "
                         "bound).")

@@ -520,7 +520,7 @@ def train(flags):  # noqa: C901
         rollout_budget_mb=flags.rollout_buffer_budget_mb,
         envs_per_thread=(
             flags.envs_per_thread if flags.envs_per_thread > 0
-            else (8 if tbflags.parse_synthetic_env_spec(flags.env) is not None
+            else (16 if tbflags.parse_synthetic_env_spec(flags.env) is not None
                   else 1)),
     )
 
