@@ -1,5 +1,13 @@
 """Deep IMPALA ResNet (ref: torchbeast/polybeast_learner.py:134-266).
 
+On GPU the 84x84 trunk runs on hand-written MFMA kernels: every 3x3 s1 p1
+conv after the first goes through ops.functional.resnet_conv3x3 (bf16
+channels_last activations end to end; conv_mfma.hip MODE 5/3 + wgrad),
+with ReLU / max-pool / residual adds on bf16 channels_last ATen ops. The
+first conv (CI = observation channels, e.g. 4) stays on ATen — its K-runs
+are narrower than an MFMA A-fragment. Features are flattened in logical
+NCHW order in both paths so checkpoints transfer between CPU and GPU.
+
 Three feature sections with channel widths [16, 32, 32]; each section is a
 3x3 s1 p1 conv, a 3x3 s2 p1 max-pool, and two residual blocks of
 (ReLU, conv3x3, ReLU, conv3x3) with identity skip. For 84x84x4 inputs the
@@ -9,6 +17,8 @@ reference deep net); optional single-layer LSTM with hidden size 256 and
 done-masked state resets. Returns a (action, policy_logits, baseline) tuple
 like the reference polybeast net.
 """
+
+import os
 
 import torch
 from torch import nn
@@ -69,6 +79,36 @@ class ResNet(nn.Module):
         self.policy = nn.Linear(core_size, num_actions)
         self.baseline = nn.Linear(core_size, 1)
 
+    def _features_mfma(self, x):
+        """84x84 trunk on the MFMA conv kernels, bf16 channels_last."""
+        from torchbeast_amd.ops import functional as tbf
+
+        cl = torch.channels_last
+
+        def res_block(blk, x):
+            out = tbf.resnet_conv3x3(blk.conv0, F.relu(x))
+            out = tbf.resnet_conv3x3(blk.conv1, F.relu(out))
+            return out + x
+
+        s1, s2, s3 = self.feat_extract
+        x = F.max_pool2d(s1.conv(x), 3, 2, 1)  # fp32 ATen (CI=obs channels)
+        x = x.to(torch.bfloat16).contiguous(memory_format=cl)
+        x = res_block(s1.res1, res_block(s1.res0, x))
+        x = F.max_pool2d(tbf.resnet_conv3x3(s2.conv, x), 3, 2, 1)
+        x = res_block(s2.res1, res_block(s2.res0, x))
+        x = F.max_pool2d(tbf.resnet_conv3x3(s3.conv, x), 3, 2, 1)
+        x = res_block(s3.res1, res_block(s3.res0, x))
+        return x.float()
+
+    def _features(self, x):
+        if (x.is_cuda and x.shape[2:] == (84, 84)
+                and os.environ.get("TBAMD_RESNET") != "aten"):
+            from torchbeast_amd.ops import functional as tbf
+
+            if tbf.resnet_conv3x3_supported(16, 42, 16):
+                return self._features_mfma(x)
+        return self.feat_extract(x)
+
     def initial_state(self, batch_size=1):
         if not self.use_lstm:
             return tuple()
@@ -81,8 +121,10 @@ class ResNet(nn.Module):
         frame, reward, done = inputs["frame"], inputs["reward"], inputs["done"]
         T, B = frame.shape[:2]
         x = torch.flatten(frame, 0, 1).float() / 255.0
-        x = self.feat_extract(x)
-        x = F.relu(self.fc(F.relu(x).view(T * B, -1)))
+        x = self._features(x)
+        # reshape (not view): the MFMA path returns channels_last; logical
+        # NCHW flatten order is preserved either way.
+        x = F.relu(self.fc(F.relu(x).reshape(T * B, -1)))
 
         clipped_reward = torch.clamp(reward, -1, 1).view(T * B, 1)
         core_input = torch.cat([x, clipped_reward], dim=-1)

@@ -374,6 +374,80 @@ def atari_trunk(frames, conv1, conv2, conv3):
 
 
 # ---------------------------------------------------------------------------
+# Deep-ResNet 3x3 s1 p1 conv on the same MFMA implicit-GEMM template
+# (activations stay bf16 channels_last through the trunk; see
+# models/resnet.py, which mirrors the reference deep net of
+# polybeast_learner.py:134-266).
+# ---------------------------------------------------------------------------
+
+
+def _pack_resnet_weight(w, rotate=False):
+    """[CO,CI,3,3] fp32 -> bf16 [rows, KP] (ky,kx,c)-major, K zero-padded to
+    a multiple of 32 (the kernel's MFMA K-step). rotate=True builds the
+    dgrad operand [CI, ky,kx,CO] with spatially flipped taps."""
+    if rotate:
+        w = w.detach().flip(2, 3).permute(1, 2, 3, 0)
+    else:
+        w = w.detach().permute(0, 2, 3, 1)
+    rows = w.shape[0]
+    flat = w.reshape(rows, -1).to(torch.bfloat16)
+    k = flat.shape[1]
+    kp = (k + 31) // 32 * 32
+    if kp != k:
+        flat = torch.cat([flat, flat.new_zeros(rows, kp - k)], dim=1)
+    return flat.contiguous()
+
+
+class _ResNetConv3x3(torch.autograd.Function):
+    """3x3 s1 p1 conv over bf16 channels_last activations, MFMA fwd/dgrad
+    (conv_mfma.hip conv_nhwc_kernel MODE 5/3) + transposed-GEMM wgrad."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        x = x.contiguous(memory_format=torch.channels_last)
+        ci, hw, co = x.shape[1], x.shape[2], weight.shape[0]
+        ext = _ext_for(x, "resnet_conv")
+        wp = _pack_resnet_weight(weight)
+        out = ext.resnet_conv(x, wp, bias.detach().float().contiguous(),
+                              ci, hw, co, True)
+        ctx.save_for_backward(x, weight)
+        ctx.geom = (ci, hw, co)
+        # NHWC-dense result presented as logical NCHW (channels_last view).
+        return out.permute(0, 3, 1, 2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        ci, hw, co = ctx.geom
+        ext = ops_mod.require_ext()
+        dyn = (dy.to(torch.bfloat16)
+               .contiguous(memory_format=torch.channels_last))
+        wr = _pack_resnet_weight(weight, rotate=True)
+        dx = ext.resnet_conv(dyn, wr, dyn.new_empty(0).float(),
+                             co, hw, ci, False).permute(0, 3, 1, 2)
+        dwp, db = ext.resnet_conv_wgrad(x, dyn, ci, hw, co)
+        # [ky][co][kx*ci+c] (K-cols padded to 16) -> [co][ci][kh][kw].
+        dw = (dwp[:, :, :3 * ci].view(3, co, 3, ci)
+              .permute(1, 3, 0, 2).contiguous())
+        return dx, dw, db
+
+
+def resnet_conv3x3(conv, x):
+    """Apply an nn.Conv2d(3x3, s1, p1) via the MFMA kernel; x is bf16
+    channels_last on GPU with a supported geometry (checked by caller)."""
+    return _ResNetConv3x3.apply(x, conv.weight, conv.bias)
+
+
+def resnet_conv3x3_supported(ci, hw, co):
+    try:
+        import torchbeast_amd.ops as _ops
+        ext = _ops.require_ext()
+    except Exception:  # noqa: BLE001
+        return False
+    return bool(ext.resnet_conv_supported(ci, hw, co))
+
+
+# ---------------------------------------------------------------------------
 # Policy sampling: softmax + multinomial (train) / argmax (eval).
 # ---------------------------------------------------------------------------
 

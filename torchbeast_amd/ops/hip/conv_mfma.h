@@ -41,4 +41,18 @@ std::vector<torch::Tensor> conv_trunk_wgrad2(torch::Tensor a1,
 std::vector<torch::Tensor> conv_trunk_wgrad3(torch::Tensor a2,
                                              torch::Tensor d3m);
 
+// Deep IMPALA ResNet 3x3 s1 p1 convs over bf16 channels_last activations
+// (reference models/resnet_monobeast.py feat_extract; see models/resnet.py).
+// x: bf16 NHWC storage [N,CI,HW,HW]; w: bf16 [CO, ceil32(9*CI)] packed
+// (ky,kx,c)-major, zero-padded; bias fp32 [CO] for fwd (empty for dgrad,
+// which takes pre-rotated weights). Returns bf16 NHWC [N,CO,HW,HW].
+torch::Tensor resnet_conv(torch::Tensor x, torch::Tensor w,
+                          torch::Tensor bias, int64_t ci, int64_t hw,
+                          int64_t co, bool fwd);
+bool resnet_conv_supported(int64_t ci, int64_t hw, int64_t co);
+// {dW [3, CO, KWCP] fp32, db [CO] fp32}; KWCP = ceil16(3*CI).
+std::vector<torch::Tensor> resnet_conv_wgrad(torch::Tensor x,
+                                             torch::Tensor dy, int64_t ci,
+                                             int64_t hw, int64_t co);
+
 }  // namespace tbamd

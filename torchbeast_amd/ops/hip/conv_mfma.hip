@@ -242,6 +242,7 @@ __global__ __launch_bounds__(NT) void conv_nhwc_kernel(
   constexpr int ROWE = IWL * CI;           // elements per logical row
   constexpr int LROWS = (OYT - 1) * ST + KH;  // staged logical rows
   constexpr int K = KH * KW * CI;
+  constexpr int KP = ((K + 31) / 32) * 32;  // W is packed zero-padded to KP
   constexpr int M_BLK = SB * OYT * OW;
   constexpr int MF = (M_BLK + 15) / 16;
   constexpr int NF = CO / 16;
@@ -249,7 +250,11 @@ __global__ __launch_bounds__(NT) void conv_nhwc_kernel(
   constexpr int MAX_MF = (MF + NW - 1) / NW;
   constexpr int BANDS = (OH + OYT - 1) / OYT;
   static_assert(ROWE % 8 == 0, "LDS rows must be whole 16 B chunks");
-  static_assert((KW * CI) % 32 == 0, "K-steps must not cross ky rows");
+  // A-fragments read 8 contiguous (kx*CI+c) elements: runs must not cross
+  // a ky row. When KW*CI is not a multiple of 32, a 32-wide K-step spans
+  // two ky rows and ky becomes per-lane-group instead of wave-uniform.
+  static_assert((KW * CI) % 8 == 0, "K-runs must not cross ky rows");
+  constexpr bool kKyUniform = ((KW * CI) % 32) == 0;
   static_assert(CI % 8 == 0 || (PAD == 0 && DIL == 1),
                 "pad/dilate staging assumes chunks within one x");
   static_assert(SB == 1 || OYT == OH, "multi-sample blocks stage full rows");
@@ -319,17 +324,24 @@ __global__ __launch_bounds__(NT) void conv_nhwc_kernel(
     for (int j = 0; j < NF; ++j) acc[f][j] = {0.f, 0.f, 0.f, 0.f};
 
 #pragma unroll
-  for (int ks = 0; ks < K / 32; ++ks) {
+  for (int ks = 0; ks < KP / 32; ++ks) {
     const int k0 = ks * 32;
-    const int ky = k0 / (KW * CI);   // wave-uniform: K-steps never cross ky
-    const int kin = k0 % (KW * CI) + 8 * lg;
     bf16x8 bfr[NF];
 #pragma unroll
     for (int j = 0; j < NF; ++j) {
       bfr[j] = *reinterpret_cast<const bf16x8*>(
-          &W[(int64_t)(j * 16 + ln) * K + k0 + 8 * lg]);
+          &W[(int64_t)(j * 16 + ln) * KP + k0 + 8 * lg]);
     }
-    const int rowoff = ky * ROWE + kin;
+    int rowoff;
+    if (kKyUniform) {
+      const int ky = k0 / (KW * CI);
+      rowoff = ky * ROWE + k0 % (KW * CI) + 8 * lg;
+    } else {
+      // Per-lane-group ky; K-pad lanes read LDS offset 0 (finite) and the
+      // zero-padded W columns annihilate the products.
+      const int k = k0 + 8 * lg;
+      rowoff = (KP == K || k < K) ? (k / (KW * CI)) * ROWE + k % (KW * CI) : 0;
+    }
 #pragma unroll
     for (int f = 0; f < MAX_MF; ++f) {
       if (f >= nmf) break;
@@ -348,7 +360,7 @@ __global__ __launch_bounds__(NT) void conv_nhwc_kernel(
 #pragma unroll
     for (int j = 0; j < NF; ++j) {
       const int cch = j * 16 + ln;
-      const float b = (MODE == 2) ? 0.f : bias[cch];
+      const float b = (MODE == 2 || MODE == 3) ? 0.f : bias[cch];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int m = f * 16 + lg * 4 + r;
@@ -367,7 +379,7 @@ __global__ __launch_bounds__(NT) void conv_nhwc_kernel(
           reinterpret_cast<float*>(out)[s * (CO * OH * OW) + cch * (OH * OW) +
                                         oy * OW + ox] = v;
         } else {
-          v = v > 0.f ? v : 0.f;
+          if (MODE == 0) v = v > 0.f ? v : 0.f;  // 3: plain dgrad, 5: +bias
           reinterpret_cast<__bf16*>(out)[((s * OH + oy) * OW + ox) * CO + cch] =
               (__bf16)v;
         }
@@ -407,7 +419,7 @@ __global__ void mask_d3_kernel(const float* __restrict__ dflat,
 // ---------------------------------------------------------------------------
 
 template <int CI, int XH, int XW, int KH, int KW, int ST, int CO, int OH,
-          int OW, int MC, bool X_U8,
+          int OW, int MC, bool X_U8, int PAD = 0,
           int KWCP = ((KW * CI + 15) / 16) * 16>
 __global__ __launch_bounds__(kThreads) void wgrad_kernel(
     const void* __restrict__ xin,    // NHWC bf16 [N,XH,XW,CI] or u8 NCHW
@@ -499,8 +511,22 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
           const int64_t s = m / (OH * OW);
           const int rm = (int)(m % (OH * OW));
           const int oy = rm / OW, ox = rm % OW;
-          v = *reinterpret_cast<const bf16x8*>(
-              &xb[((s * XH + oy * ST + ky) * XW + ox * ST) * CI + ch]);
+          const int iy = oy * ST + ky - PAD;
+          if (PAD == 0) {
+            v = *reinterpret_cast<const bf16x8*>(
+                &xb[((s * XH + iy) * XW + ox * ST) * CI + ch]);
+          } else if (iy >= 0 && iy < XH) {
+            // Padded convs: the 8-element chunk may straddle the x border;
+            // guard per element (staging only, not the hot K-loop).
+#pragma unroll
+            for (int i = 0; i < 8; ++i) {
+              const int e = ch + i;           // kx*CI + c within the span
+              const int ix = ox * ST + e / CI - PAD;
+              if (ix >= 0 && ix < XW) {
+                v[i] = xb[((s * XH + iy) * XW + ix) * CI + e % CI];
+              }
+            }
+          }
         }
 #pragma unroll
         for (int i = 0; i < 8; ++i) sXT[ch + i][mm] = v[i];
@@ -820,7 +846,7 @@ torch::Tensor conv_trunk_dgrad2(torch::Tensor d2, torch::Tensor w2r,
 namespace {
 
 template <int CI, int XH, int XW, int KH, int KW, int ST, int CO, int OH,
-          int OW, int MC, bool X_U8>
+          int OW, int MC, bool X_U8, int PAD = 0>
 std::vector<torch::Tensor> run_wgrad(const torch::Tensor& x,
                                      const torch::Tensor& dy) {
   constexpr int KWC = ((KW * CI + 15) / 16) * 16;  // padded (kernel KWCP)
@@ -836,7 +862,7 @@ std::vector<torch::Tensor> run_wgrad(const torch::Tensor& x,
   auto db = torch::empty({CO}, fopt);
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(
-      (wgrad_kernel<CI, XH, XW, KH, KW, ST, CO, OH, OW, MC, X_U8>),
+      (wgrad_kernel<CI, XH, XW, KH, KW, ST, CO, OH, OW, MC, X_U8, PAD>),
       dim3(KH, nchunks), dim3(kThreads), 0, stream,
       x.data_ptr(), reinterpret_cast<const __bf16*>(dy.data_ptr()),
       partials.data_ptr<float>(), db_partials.data_ptr<float>(), N);
@@ -900,6 +926,111 @@ std::vector<torch::Tensor> conv_trunk_wgrad3(torch::Tensor a2,
                      F3::OH, F3::OW, 1024, false>(a2, d3m);
   }
   TORCH_CHECK(false, "wgrad3: unsupported geometry ", a2.sizes());
+}
+
+// ---------------------------------------------------------------------------
+// Deep IMPALA ResNet 3x3 s1 p1 convs (bf16 channels_last activations).
+// Geometry table covers the 84x84x4 deep net: sections at 42x42 (16ch),
+// 21x21 and 11x11 (32ch). The first conv (4->16 @84) stays on library ops
+// (C=4 NHWC runs are narrower than an A-fragment).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <int CI_, int HW, int CO_, int OYT_>
+struct RGeom {
+  static constexpr int CI = CI_, CO = CO_;
+  static constexpr int IHR = HW, IWR = HW, PAD = 1, DIL = 1;
+  static constexpr int KH = 3, KW = 3, ST = 1;
+  static constexpr int OH = HW, OW = HW, SB = 1, OYT = OYT_;
+  static constexpr int IWL = HW + 2;
+  static constexpr size_t LDS = (size_t)((OYT - 1) * ST + KH) * IWL * CI * 2;
+};
+
+using RC_16_42_16 = RGeom<16, 42, 16, 7>;   // section-1 residual convs
+using RC_16_42_32 = RGeom<16, 42, 32, 7>;   // section-2 feature conv
+using RC_32_42_16 = RGeom<32, 42, 16, 7>;   // ... and its dgrad
+using RC_32_21_32 = RGeom<32, 21, 32, 7>;   // section-2 res / section-3 feat
+using RC_32_11_32 = RGeom<32, 11, 32, 11>;  // section-3 residual convs
+
+template <typename G, int MODE>
+torch::Tensor launch_rconv(const torch::Tensor& in, const torch::Tensor& w,
+                           const float* bias_p, int64_t N) {
+  auto out = torch::empty({N, G::OH, G::OW, G::CO}, w.options());
+  constexpr int kBands = (G::OH + G::OYT - 1) / G::OYT;
+  hipLaunchKernelGGL(
+      (conv_nhwc_kernel<G::CI, G::IHR, G::IWR, G::PAD, G::DIL, G::KH, G::KW,
+                        G::ST, G::CO, G::OH, G::OW, G::SB, G::OYT, MODE>),
+      dim3((uint32_t)(N * kBands)), dim3(kThreads), G::LDS,
+      at::cuda::getCurrentCUDAStream(),
+      reinterpret_cast<const __bf16*>(in.data_ptr()),
+      reinterpret_cast<const __bf16*>(w.data_ptr()), bias_p, nullptr,
+      out.data_ptr(), (int)N);
+  return out;
+}
+
+}  // namespace
+
+// x: bf16 channels_last [N, CI, HW, HW] (NHWC storage); w: bf16
+// [CO, KP] packed (ky,kx,c)-major zero-padded to KP = ceil32(9*CI);
+// bias fp32 [CO] for fwd, empty for dgrad (which takes pre-rotated w).
+torch::Tensor resnet_conv(torch::Tensor x, torch::Tensor w,
+                          torch::Tensor bias, int64_t ci, int64_t hw,
+                          int64_t co, bool fwd) {
+  check_bf16(w, "resnet w");
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16,
+              "resnet_conv: bf16 GPU input required");
+  const int64_t N = x.numel() / (ci * hw * hw);
+  const float* bias_p =
+      bias.numel() > 0 ? bias.data_ptr<float>() : nullptr;
+  TORCH_CHECK(fwd == (bias_p != nullptr), "fwd needs bias, dgrad must not");
+  if (ci == 16 && hw == 42 && co == 16) {
+    return fwd ? launch_rconv<RC_16_42_16, 5>(x, w, bias_p, N)
+               : launch_rconv<RC_16_42_16, 3>(x, w, bias_p, N);
+  }
+  if (ci == 16 && hw == 42 && co == 32) {
+    return launch_rconv<RC_16_42_32, 5>(x, w, bias_p, N);
+  }
+  if (ci == 32 && hw == 42 && co == 16) {
+    return launch_rconv<RC_32_42_16, 3>(x, w, bias_p, N);
+  }
+  if (ci == 32 && hw == 21 && co == 32) {
+    return fwd ? launch_rconv<RC_32_21_32, 5>(x, w, bias_p, N)
+               : launch_rconv<RC_32_21_32, 3>(x, w, bias_p, N);
+  }
+  if (ci == 32 && hw == 11 && co == 32) {
+    return fwd ? launch_rconv<RC_32_11_32, 5>(x, w, bias_p, N)
+               : launch_rconv<RC_32_11_32, 3>(x, w, bias_p, N);
+  }
+  TORCH_CHECK(false, "resnet_conv: unsupported geometry ci=", ci, " hw=", hw,
+              " co=", co);
+}
+
+bool resnet_conv_supported(int64_t ci, int64_t hw, int64_t co) {
+  return (ci == 16 && hw == 42 && (co == 16 || co == 32)) ||
+         (ci == 32 && hw == 42 && co == 16) ||
+         (ci == 32 && hw == 21 && co == 32) ||
+         (ci == 32 && hw == 11 && co == 32);
+}
+
+// wgrad for the 3x3 s1 p1 convs; x/dy bf16 NHWC storage. Returns
+// {dW [3, CO, KWCP] fp32, db [CO] fp32}.
+std::vector<torch::Tensor> resnet_conv_wgrad(torch::Tensor x,
+                                             torch::Tensor dy, int64_t ci,
+                                             int64_t hw, int64_t co) {
+  if (ci == 16 && hw == 42 && co == 16) {
+    return run_wgrad<16, 42, 42, 3, 3, 1, 16, 42, 42, 2048, false, 1>(x, dy);
+  }
+  if (ci == 16 && hw == 42 && co == 32) {
+    return run_wgrad<16, 42, 42, 3, 3, 1, 32, 42, 42, 2048, false, 1>(x, dy);
+  }
+  if (ci == 32 && hw == 21 && co == 32) {
+    return run_wgrad<32, 21, 21, 3, 3, 1, 32, 21, 21, 2048, false, 1>(x, dy);
+  }
+  if (ci == 32 && hw == 11 && co == 32) {
+    return run_wgrad<32, 11, 11, 3, 3, 1, 32, 11, 11, 1024, false, 1>(x, dy);
+  }
+  TORCH_CHECK(false, "resnet_conv_wgrad: unsupported geometry");
 }
 
 }  // namespace tbamd

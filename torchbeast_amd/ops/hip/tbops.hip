@@ -536,6 +536,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_trunk_wgrad1", &tbamd::conv_trunk_wgrad1);
   m.def("conv_trunk_wgrad2", &tbamd::conv_trunk_wgrad2);
   m.def("conv_trunk_wgrad3", &tbamd::conv_trunk_wgrad3);
+  m.def("resnet_conv", &tbamd::resnet_conv);
+  m.def("resnet_conv_supported", &tbamd::resnet_conv_supported);
+  m.def("resnet_conv_wgrad", &tbamd::resnet_conv_wgrad);
   m.def("barrier_probe", &tbamd::barrier_probe);
   m.def("launch_probe", &tbamd::launch_probe);
   m.def("mfma_gemm", &tbamd::mfma_gemm);
