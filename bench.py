@@ -256,11 +256,12 @@ def main():
 
     graph = None
     static_batch = None
-    # Default hipGraph only for the shallow non-LSTM headline config: the
-    # LSTM's cooperative persistent kernels capture but replay ~3x slower,
-    # and the deep model sees no benefit (measured).
+    # Default hipGraph for the non-LSTM configs (the LSTM's cooperative
+    # persistent kernels capture but replay ~3x slower). The deep model's
+    # ~150-op trunk step gains the most: replay removes the per-op GIL
+    # handoffs that actor threads otherwise inflate to >100 ms/step.
     use_graph = (args.hipgraph and use_cuda and world_size == 1
-                 and args.model == "shallow" and not args.use_lstm)
+                 and not args.use_lstm)
 
     # Warmup (fills the pipeline, compiles/caches kernels).
     for _ in range(args.warmup):

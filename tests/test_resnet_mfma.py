@@ -38,6 +38,29 @@ def test_resnet_conv_fwd(ci, hw, co, n):
     torch.testing.assert_close(out.float(), ref, rtol=5e-2, atol=2e-2)
 
 
+def test_resnet_first_conv_fwd_and_wgrad():
+    # Obs channels zero-padded to 8; backward is wgrad-only.
+    torch.manual_seed(11)
+    conv = torch.nn.Conv2d(4, 16, 3, padding=1).cuda()
+    x = torch.rand(7, 4, 84, 84, device="cuda")
+    out = tbf.resnet_first_conv(conv, x)
+    with torch.no_grad():
+        ref = F.conv2d(_bf(x), _bf(conv.weight), conv.bias, padding=1)
+    torch.testing.assert_close(out.float(), ref, rtol=5e-2, atol=2e-2)
+
+    dy = torch.randn_like(out)
+    out.backward(dy)
+    from torch.nn.grad import conv2d_weight
+    dyb = _bf(dy.float())
+    dw_ref = conv2d_weight(_bf(x), conv.weight.shape, dyb, padding=1)
+    db_ref = dyb.sum((0, 2, 3))
+    for got, ref_g, name in [(conv.weight.grad, dw_ref, "dw"),
+                             (conv.bias.grad, db_ref, "db")]:
+        scale = ref_g.abs().max().clamp_min(1e-4)
+        err = (got.float() - ref_g).abs().max() / scale
+        assert err < 2e-2, f"{name}: rel-max err {err:.4f}"
+
+
 def test_resnet_dgrad_geometry_42():
     # The section-2 feature conv's dgrad runs the 32->16 @42 geometry.
     torch.manual_seed(7)

@@ -91,8 +91,10 @@ class ResNet(nn.Module):
             return out + x
 
         s1, s2, s3 = self.feat_extract
-        x = F.max_pool2d(s1.conv(x), 3, 2, 1)  # fp32 ATen (CI=obs channels)
-        x = x.to(torch.bfloat16).contiguous(memory_format=cl)
+        # First conv: obs channels zero-padded to 8 on the same MFMA
+        # template (wgrad-only backward; frames carry no grad).
+        x = F.max_pool2d(tbf.resnet_first_conv(s1.conv, x), 3, 2, 1)
+        x = x.contiguous(memory_format=cl)
         x = res_block(s1.res1, res_block(s1.res0, x))
         x = F.max_pool2d(tbf.resnet_conv3x3(s2.conv, x), 3, 2, 1)
         x = res_block(s2.res1, res_block(s2.res0, x))
@@ -101,7 +103,7 @@ class ResNet(nn.Module):
         return x.float()
 
     def _features(self, x):
-        if (x.is_cuda and x.shape[2:] == (84, 84)
+        if (x.is_cuda and x.shape[2:] == (84, 84) and x.shape[1] <= 8
                 and os.environ.get("TBAMD_RESNET") != "aten"):
             from torchbeast_amd.ops import functional as tbf
 

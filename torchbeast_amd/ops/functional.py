@@ -438,6 +438,51 @@ def resnet_conv3x3(conv, x):
     return _ResNetConv3x3.apply(x, conv.weight, conv.bias)
 
 
+class _ResNetFirstConv(torch.autograd.Function):
+    """First deep-net conv (obs channels -> 16 @84): input channels are
+    zero-padded to 8 so the K-runs fill MFMA A-fragments. Frames carry no
+    grad, so backward is wgrad-only (the zero-channel taps are sliced
+    off). Replaces MIOpen's fp32 NCHW wrw, which falls back to a naive
+    double-accumulation kernel on this geometry."""
+
+    @staticmethod
+    def forward(ctx, x8, weight, bias):
+        co, ci = weight.shape[0], weight.shape[1]
+        w8 = torch.cat(
+            [weight.detach(),
+             weight.new_zeros(co, 8 - ci, *weight.shape[2:])], dim=1)
+        ext = _ext_for(x8, "resnet_conv")
+        out = ext.resnet_conv(x8, _pack_resnet_weight(w8),
+                              bias.detach().float().contiguous(),
+                              8, x8.shape[2], co, True)
+        ctx.save_for_backward(x8)
+        ctx.ci = ci
+        return out.permute(0, 3, 1, 2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x8,) = ctx.saved_tensors
+        ext = ops_mod.require_ext()
+        dyn = (dy.to(torch.bfloat16)
+               .contiguous(memory_format=torch.channels_last))
+        co = dyn.shape[1]
+        dwp, db = ext.resnet_conv_wgrad(x8, dyn, 8, x8.shape[2], co)
+        dw = (dwp[:, :, :24].view(3, co, 3, 8).permute(1, 3, 0, 2)
+              [:, :ctx.ci].contiguous())
+        return None, dw, db
+
+
+def resnet_first_conv(conv, x):
+    """x: fp32 [N, C<=8, 84, 84] frames (no grad); returns bf16
+    channels_last conv output."""
+    n, c = x.shape[0], x.shape[1]
+    x8 = torch.empty((n, 8, *x.shape[2:]), dtype=torch.bfloat16,
+                     device=x.device, memory_format=torch.channels_last)
+    x8.zero_()
+    x8[:, :c] = x
+    return _ResNetFirstConv.apply(x8, conv.weight, conv.bias)
+
+
 def resnet_conv3x3_supported(ci, hw, co):
     try:
         import torchbeast_amd.ops as _ops

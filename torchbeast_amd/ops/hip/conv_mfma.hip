@@ -947,6 +947,9 @@ struct RGeom {
   static constexpr size_t LDS = (size_t)((OYT - 1) * ST + KH) * IWL * CI * 2;
 };
 
+using RC_8_84_16 = RGeom<8, 84, 16, 7>;     // first conv, obs channels
+                                            // zero-padded 4->8 (no dgrad:
+                                            // frames carry no grad)
 using RC_16_42_16 = RGeom<16, 42, 16, 7>;   // section-1 residual convs
 using RC_16_42_32 = RGeom<16, 42, 32, 7>;   // section-2 feature conv
 using RC_32_42_16 = RGeom<32, 42, 16, 7>;   // ... and its dgrad
@@ -984,6 +987,10 @@ torch::Tensor resnet_conv(torch::Tensor x, torch::Tensor w,
   const float* bias_p =
       bias.numel() > 0 ? bias.data_ptr<float>() : nullptr;
   TORCH_CHECK(fwd == (bias_p != nullptr), "fwd needs bias, dgrad must not");
+  if (ci == 8 && hw == 84 && co == 16) {
+    TORCH_CHECK(fwd, "first conv has no dgrad (frames carry no grad)");
+    return launch_rconv<RC_8_84_16, 5>(x, w, bias_p, N);
+  }
   if (ci == 16 && hw == 42 && co == 16) {
     return fwd ? launch_rconv<RC_16_42_16, 5>(x, w, bias_p, N)
                : launch_rconv<RC_16_42_16, 3>(x, w, bias_p, N);
@@ -1007,7 +1014,8 @@ torch::Tensor resnet_conv(torch::Tensor x, torch::Tensor w,
 }
 
 bool resnet_conv_supported(int64_t ci, int64_t hw, int64_t co) {
-  return (ci == 16 && hw == 42 && (co == 16 || co == 32)) ||
+  return (ci == 8 && hw == 84 && co == 16) ||
+         (ci == 16 && hw == 42 && (co == 16 || co == 32)) ||
          (ci == 32 && hw == 42 && co == 16) ||
          (ci == 32 && hw == 21 && co == 32) ||
          (ci == 32 && hw == 11 && co == 32);
@@ -1018,6 +1026,9 @@ bool resnet_conv_supported(int64_t ci, int64_t hw, int64_t co) {
 std::vector<torch::Tensor> resnet_conv_wgrad(torch::Tensor x,
                                              torch::Tensor dy, int64_t ci,
                                              int64_t hw, int64_t co) {
+  if (ci == 8 && hw == 84 && co == 16) {
+    return run_wgrad<8, 84, 84, 3, 3, 1, 16, 84, 84, 4096, false, 1>(x, dy);
+  }
   if (ci == 16 && hw == 42 && co == 16) {
     return run_wgrad<16, 42, 42, 3, 3, 1, 16, 42, 42, 2048, false, 1>(x, dy);
   }
