@@ -152,8 +152,14 @@ class InferenceRunner {
     // been observed to fault; once found, the solution cache makes later
     // serves safe to run fully in parallel.
     const int64_t bq = (batch.size() + 63) / 64 * 64;
+    // Deep 84x84 trunk serves on the hand-written MFMA 3x3 kernels
+    // (conv_mfma.hip resnet_conv); MIOpen only for other geometries.
+    const bool deep_mfma =
+        deep_ && !aten_only_ && weights_[0].size(0) == 16 &&
+        weights_[0].size(1) <= 8 && weights_[0].size(2) == 3 &&
+        tbamd::resnet_conv_supported(16, 42, 16);
     std::unique_lock<std::mutex> warm_lock;
-    if (deep_) {
+    if (deep_ && !deep_mfma) {
       // Only the MIOpen (deep ResNet) path needs first-serve-per-shape
       // serialization: concurrent solution-finds for a new conv shape
       // across streams have been observed to fault. The hand-written
@@ -215,7 +221,61 @@ class InferenceRunner {
     }
 
     torch::Tensor x;
-    if (deep_) {
+    if (deep_ && deep_mfma && H == 84 && W == 84 && C <= 8) {
+      // bf16 channels_last trunk, every conv on the MFMA template. Packed
+      // bf16 weights cached until the learner's next sync.
+      std::vector<torch::Tensor> wp;
+      {
+        std::lock_guard<std::mutex> g(pack_mu_);
+        const int64_t v = weights_version_.load();
+        if (deep_packed_version_ != v) {
+          deep_packed_.clear();
+          for (size_t wi = 0; wi < 30; wi += 2) {
+            torch::Tensor w = weights_[wi];
+            if (wi == 0 && w.size(1) < 8) {  // first conv: pad channels
+              w = torch::cat(
+                  {w, w.new_zeros({w.size(0), 8 - w.size(1), 3, 3})}, 1);
+            }
+            auto f = w.permute({0, 2, 3, 1}).reshape({w.size(0), -1})
+                         .to(torch::kBFloat16);
+            const int64_t k = f.size(1), kpad = (k + 31) / 32 * 32 - k;
+            if (kpad) f = torch::cat({f, f.new_zeros({f.size(0), kpad})}, 1);
+            deep_packed_.push_back(f.contiguous());
+          }
+          stream.synchronize();
+          deep_packed_version_ = v;
+        }
+        wp = deep_packed_;
+      }
+      auto x8 = torch::empty({bp, 8, 84, 84}, opts.dtype(torch::kBFloat16),
+                             torch::MemoryFormat::ChannelsLast);
+      x8.zero_();
+      x8.narrow(1, 0, C).copy_(frames_p, /*non_blocking=*/true);
+      x8.mul_(1.0f / 255.0f);
+      size_t wi = 0, pi = 0;
+      auto conv = [&](torch::Tensor t, int64_t ci, int64_t hw, int64_t co) {
+        auto o = tbamd::resnet_conv(t, wp[pi], weights_[wi + 1], ci, hw, co,
+                                    /*fwd=*/true);
+        ++pi;
+        wi += 2;
+        return o.permute({0, 3, 1, 2});
+      };
+      const int64_t widths[3] = {16, 32, 32};
+      int64_t hw = 84, ci = 8;
+      torch::Tensor t = x8;
+      for (int sec = 0; sec < 3; ++sec) {
+        t = conv(t, ci, hw, widths[sec]);
+        t = at::max_pool2d(t, 3, 2, 1);
+        hw = (hw + 1) / 2;
+        ci = widths[sec];
+        for (int res = 0; res < 2; ++res) {
+          torch::Tensor y = conv(at::relu(t), ci, hw, ci);
+          y = conv(at::relu(y), ci, hw, ci);
+          t = t + y;
+        }
+      }
+      x = at::relu(t).to(torch::kFloat32).reshape({bp, -1});
+    } else if (deep_) {
       x = frames_p.to(torch::kFloat32).mul_(1.0f / 255.0f);
       size_t wi = 0;
       for (int sec = 0; sec < 3; ++sec) {
@@ -435,6 +495,8 @@ class InferenceRunner {
   std::vector<int64_t> frame_shape_;
   std::mutex pack_mu_;
   torch::Tensor w1p_, w2p_, w3p_;
+  std::vector<torch::Tensor> deep_packed_;
+  int64_t deep_packed_version_ = -1;
   int64_t packed_version_ = -1;
   std::atomic<int64_t> weights_version_{0};
   std::atomic<uint64_t> seed_ctr_{1};
