@@ -81,6 +81,8 @@ def test_deep_runner_matches_model():
     (action, logits, baseline, new_state), ref_out, _ = _serve_once(
         model, b=3, use_lstm=False
     )
-    torch.testing.assert_close(logits, ref_out[1].cpu(), rtol=1e-4, atol=1e-4)
-    torch.testing.assert_close(baseline, ref_out[2].cpu(), rtol=1e-4,
-                               atol=1e-4)
+    # Both sides run the bf16 MFMA trunk, but normalization rounding
+    # differs (model: fp32/255 -> bf16; runner: u8 -> bf16, bf16 mul).
+    torch.testing.assert_close(logits, ref_out[1].cpu(), rtol=5e-2, atol=5e-3)
+    torch.testing.assert_close(baseline, ref_out[2].cpu(), rtol=5e-2,
+                               atol=5e-3)
