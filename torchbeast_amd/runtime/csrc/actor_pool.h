@@ -9,6 +9,7 @@
 
 #pragma once
 
+#include <algorithm>
 #include <atomic>
 #include <condition_variable>
 #include <cstring>
@@ -285,8 +286,15 @@ class ActorPool {
             steps, [&total](const std::vector<torch::Tensor>& column) {
               int64_t rows = 0;
               for (const auto& t : column) rows += t.size(0);
+              // Logical row bytes (exactly what carve() will allocate).
+              // NOT stride(0): batch-split agent outputs are views whose
+              // stride spans the whole serve batch, which varies per
+              // serve — racing estimators would disagree on the total and
+              // a late, larger store would overflow the fixed slot size.
+              const int64_t row_numel =
+                  column[0].numel() / std::max<int64_t>(1, column[0].size(0));
               const int64_t bytes =
-                  rows * column[0].stride(0) * column[0].element_size();
+                  rows * row_numel * column[0].element_size();
               total += (bytes + 255) & ~int64_t(255);
               return column[0];
             });
