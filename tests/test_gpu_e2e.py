@@ -33,6 +33,21 @@ def test_polybeast_train_short_gpu(tmp_path):
     assert os.path.exists(os.path.join(str(tmp_path), "gpue2e", "model.tar"))
 
 
+def test_polybeast_train_deep_short_gpu(tmp_path):
+    # Deep IMPALA-ResNet path: MFMA 3x3 convs in both the learner autograd
+    # graph and the C++ runner's serving trunk.
+    subprocess.check_call(
+        [sys.executable, "-m", "torchbeast_amd.polybeast_learner",
+         "--env", "synthetic:4x84x84:6", "--model", "deep",
+         "--savedir", str(tmp_path),
+         "--xpid", "gpudeep", "--num_actors", "16", "--batch_size", "8",
+         "--unroll_length", "20", "--total_steps", str(8 * 20 * 4),
+         "--num_learner_threads", "1", "--num_inference_threads", "1"],
+        cwd=_repo_root(), timeout=300,
+    )
+    assert os.path.exists(os.path.join(str(tmp_path), "gpudeep", "model.tar"))
+
+
 def test_bench_smoke():
     out = subprocess.check_output(
         [sys.executable, "bench.py", "--gpus", "1", "--steps", "5",
