@@ -428,15 +428,22 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
     float* __restrict__ db_partials,  // [nchunks, CO]
     int N) {
   constexpr int KWC = KW * CI;  // real K-columns; KWCP pads to MFMA tiles
-  constexpr int MPAD = 40;  // 32 + 8: pads the transposed tiles' rows
+  // 128 B staging rows + the conv kernel's XOR swizzle: the scalar
+  // transpose writes (a column per lane) and the strided fragment reads
+  // otherwise serialize on the 32 LDS banks (measured ~1 conflict per
+  // VALU op at MPAD=40, profiles/trunk_mfma_pmc_r2.md).
+  constexpr int MPAD = 64;
   constexpr int OF = (CO / 16) * (KWCP / 16);
   constexpr int PER_WAVE = (OF + kWaves - 1) / kWaves;
   static_assert(KWC % 8 == 0, "KW*CI must be whole 16 B chunks");
   static_assert(KWCP % 16 == 0 && KWCP >= KWC, "bad KWCP");
   static_assert(CO % 16 == 0, "CO must tile by 16");
 
-  __shared__ __bf16 sDYT[CO][MPAD];
-  __shared__ __bf16 sXT[KWCP][MPAD];
+  __shared__ char sDYT[CO * MPAD * 2];
+  __shared__ char sXT[KWCP * MPAD * 2];
+  auto stage = [](char* buf, int row, int col, __bf16 v) {
+    *reinterpret_cast<__bf16*>(buf + swz((row * MPAD + col) * 2)) = v;
+  };
 
   const int ky = blockIdx.x;
   const int chunk = blockIdx.y;
@@ -454,7 +461,7 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
   float db = 0.f;
   // Zero the K-column pad rows once (staging never writes them).
   for (int i = tid; i < (KWCP - KWC) * MPAD; i += kThreads) {
-    sXT[KWC + i / MPAD][i % MPAD] = (__bf16)0.f;
+    stage(sXT, KWC + i / MPAD, i % MPAD, (__bf16)0.f);
   }
   if (KWCP != KWC) __syncthreads();
 
@@ -469,7 +476,7 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
         v = *reinterpret_cast<const bf16x8*>(&dy[(ms + mm) * CO + ch]);
       }
 #pragma unroll
-      for (int i = 0; i < 8; ++i) sDYT[ch + i][mm] = v[i];
+      for (int i = 0; i < 8; ++i) stage(sDYT, ch + i, mm, v[i]);
     }
     // ---- transpose-stage X spans for this ky ----
     if (X_U8) {
@@ -498,7 +505,7 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
           }
         }
 #pragma unroll
-        for (int i = 0; i < KW; ++i) sXT[c * KW + i][mm] = (__bf16)px[i];
+        for (int i = 0; i < KW; ++i) stage(sXT, c * KW + i, mm, (__bf16)px[i]);
       }
     } else {
       const __bf16* xb = reinterpret_cast<const __bf16*>(xin);
@@ -529,7 +536,7 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
           }
         }
 #pragma unroll
-        for (int i = 0; i < 8; ++i) sXT[ch + i][mm] = v[i];
+        for (int i = 0; i < 8; ++i) stage(sXT, ch + i, mm, v[i]);
       }
     }
     __syncthreads();
@@ -541,18 +548,15 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
       if (of >= OF) break;
       const int ni = of / (KWCP / 16);
       const int ki = of % (KWCP / 16);
-      const bf16x8 a =
-          *reinterpret_cast<const bf16x8*>(&sDYT[ni * 16 + ln][8 * lg]);
-      const bf16x8 b =
-          *reinterpret_cast<const bf16x8*>(&sXT[ki * 16 + ln][8 * lg]);
+      const bf16x8 a = lds_read8_swz(sDYT, (ni * 16 + ln) * MPAD + 8 * lg);
+      const bf16x8 b = lds_read8_swz(sXT, (ki * 16 + ln) * MPAD + 8 * lg);
       acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[i], 0, 0, 0);
     }
     // db: wave 0 (ky==0 blocks only; one lane per output channel).
     if (ky == 0 && wave == 0 && lane < CO) {
-      const bf16x8* row = reinterpret_cast<const bf16x8*>(&sDYT[lane][0]);
 #pragma unroll
       for (int ch = 0; ch < 4; ++ch) {
-        const bf16x8 v = row[ch];
+        const bf16x8 v = lds_read8_swz(sDYT, lane * MPAD + ch * 8);
 #pragma unroll
         for (int i = 0; i < 8; ++i) db += (float)v[i];
       }
