@@ -433,6 +433,7 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
   // otherwise serialize on the 32 LDS banks (measured ~1 conflict per
   // VALU op at MPAD=40, profiles/trunk_mfma_pmc_r2.md).
   constexpr int MPAD = 64;
+  constexpr int MSTEP = 64;  // two MFMA k-chunks per staging round
   constexpr int OF = (CO / 16) * (KWCP / 16);
   constexpr int PER_WAVE = (OF + kWaves - 1) / kWaves;
   static_assert(KWC % 8 == 0, "KW*CI must be whole 16 B chunks");
@@ -466,9 +467,9 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
   if (KWCP != KWC) __syncthreads();
 
   const int64_t mend = (m0 + MC < M) ? m0 + MC : M;
-  for (int64_t ms = m0; ms < mend; ms += 32) {
-    // ---- transpose-stage dY[ms..ms+32) ----
-    for (int idx = tid; idx < 32 * (CO / 8); idx += kThreads) {
+  for (int64_t ms = m0; ms < mend; ms += MSTEP) {
+    // ---- transpose-stage dY[ms..ms+MSTEP) ----
+    for (int idx = tid; idx < MSTEP * (CO / 8); idx += kThreads) {
       const int mm = idx / (CO / 8);
       const int ch = (idx % (CO / 8)) * 8;
       bf16x8 v = {};
@@ -482,7 +483,7 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
     if (X_U8) {
       // u8 NCHW frames; k-order (c, kx), span per (m,c) = 8 px at stride 1.
       const uint8_t* xu = reinterpret_cast<const uint8_t*>(xin);
-      for (int idx = tid; idx < 32 * CI; idx += kThreads) {
+      for (int idx = tid; idx < MSTEP * CI; idx += kThreads) {
         const int mm = idx / CI;
         const int c = idx % CI;
         const int64_t m = ms + mm;
@@ -509,7 +510,7 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
       }
     } else {
       const __bf16* xb = reinterpret_cast<const __bf16*>(xin);
-      for (int idx = tid; idx < 32 * (KWC / 8); idx += kThreads) {
+      for (int idx = tid; idx < MSTEP * (KWC / 8); idx += kThreads) {
         const int mm = idx / (KWC / 8);
         const int ch = (idx % (KWC / 8)) * 8;
         const int64_t m = ms + mm;
@@ -548,14 +549,20 @@ __global__ __launch_bounds__(kThreads) void wgrad_kernel(
       if (of >= OF) break;
       const int ni = of / (KWCP / 16);
       const int ki = of % (KWCP / 16);
-      const bf16x8 a = lds_read8_swz(sDYT, (ni * 16 + ln) * MPAD + 8 * lg);
-      const bf16x8 b = lds_read8_swz(sXT, (ki * 16 + ln) * MPAD + 8 * lg);
-      acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[i], 0, 0, 0);
+#pragma unroll
+      for (int half = 0; half < MSTEP / 32; ++half) {
+        const bf16x8 a = lds_read8_swz(
+            sDYT, (ni * 16 + ln) * MPAD + half * 32 + 8 * lg);
+        const bf16x8 b = lds_read8_swz(
+            sXT, (ki * 16 + ln) * MPAD + half * 32 + 8 * lg);
+        acc[i] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[i], 0, 0, 0);
+      }
     }
     // db: wave 0 (ky==0 blocks only; one lane per output channel).
     if (ky == 0 && wave == 0 && lane < CO) {
 #pragma unroll
-      for (int ch = 0; ch < 4; ++ch) {
+      for (int ch = 0; ch < MSTEP / 8; ++ch) {
         const bf16x8 v = lds_read8_swz(sDYT, lane * MPAD + ch * 8);
 #pragma unroll
         for (int i = 0; i < 8; ++i) db += (float)v[i];
