@@ -261,7 +261,8 @@ def main():
     # ~150-op trunk step gains the most: replay removes the per-op GIL
     # handoffs that actor threads otherwise inflate to >100 ms/step.
     use_graph = (args.hipgraph and use_cuda and world_size == 1
-                 and not args.use_lstm)
+                 and (not args.use_lstm
+                      or os.environ.get("TBAMD_GRAPH_LSTM") == "1"))
 
     # Warmup (fills the pipeline, compiles/caches kernels).
     for _ in range(args.warmup):
