@@ -941,9 +941,9 @@ std::vector<torch::Tensor> conv_trunk_wgrad3(torch::Tensor a2,
 
 // ---------------------------------------------------------------------------
 // Deep IMPALA ResNet 3x3 s1 p1 convs (bf16 channels_last activations).
-// Geometry table covers the 84x84x4 deep net: sections at 42x42 (16ch),
-// 21x21 and 11x11 (32ch). The first conv (4->16 @84) stays on library ops
-// (C=4 NHWC runs are narrower than an A-fragment).
+// Geometry table covers the 84x84x4 deep net: the first conv at 84x84
+// (obs channels zero-padded to 8 so K-runs fill an A-fragment), then
+// sections at 42x42 (16ch), 21x21 and 11x11 (32ch).
 // ---------------------------------------------------------------------------
 
 namespace {
@@ -992,12 +992,15 @@ torch::Tensor resnet_conv(torch::Tensor x, torch::Tensor w,
                           torch::Tensor bias, int64_t ci, int64_t hw,
                           int64_t co, bool fwd) {
   check_bf16(w, "resnet w");
-  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16,
-              "resnet_conv: bf16 GPU input required");
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 &&
+                  (x.dim() != 4 ||
+                   x.is_contiguous(at::MemoryFormat::ChannelsLast)),
+              "resnet_conv: channels_last bf16 GPU input required");
   const int64_t N = x.numel() / (ci * hw * hw);
   const float* bias_p =
       bias.numel() > 0 ? bias.data_ptr<float>() : nullptr;
   TORCH_CHECK(fwd == (bias_p != nullptr), "fwd needs bias, dgrad must not");
+  if (N == 0) return torch::empty({0, hw, hw, co}, w.options());
   if (ci == 8 && hw == 84 && co == 16) {
     TORCH_CHECK(fwd, "first conv has no dgrad (frames carry no grad)");
     return launch_rconv<RC_8_84_16, 5>(x, w, bias_p, N);
@@ -1007,9 +1010,11 @@ torch::Tensor resnet_conv(torch::Tensor x, torch::Tensor w,
                : launch_rconv<RC_16_42_16, 3>(x, w, bias_p, N);
   }
   if (ci == 16 && hw == 42 && co == 32) {
+    TORCH_CHECK(fwd, "16->32 @42 is the feature conv (fwd only)");
     return launch_rconv<RC_16_42_32, 5>(x, w, bias_p, N);
   }
   if (ci == 32 && hw == 42 && co == 16) {
+    TORCH_CHECK(!fwd, "32->16 @42 is the feature conv's dgrad");
     return launch_rconv<RC_32_42_16, 3>(x, w, bias_p, N);
   }
   if (ci == 32 && hw == 21 && co == 32) {
@@ -1037,6 +1042,14 @@ bool resnet_conv_supported(int64_t ci, int64_t hw, int64_t co) {
 std::vector<torch::Tensor> resnet_conv_wgrad(torch::Tensor x,
                                              torch::Tensor dy, int64_t ci,
                                              int64_t hw, int64_t co) {
+  auto check_cl = [](const torch::Tensor& t, const char* what) {
+    TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16 &&
+                    (t.dim() != 4 ||
+                     t.is_contiguous(at::MemoryFormat::ChannelsLast)),
+                what, ": channels_last bf16 GPU tensor required");
+  };
+  check_cl(x, "resnet wgrad x");
+  check_cl(dy, "resnet wgrad dy");
   if (ci == 8 && hw == 84 && co == 16) {
     return run_wgrad<8, 84, 84, 3, 3, 1, 16, 84, 84, 4096, false, 1>(x, dy);
   }
