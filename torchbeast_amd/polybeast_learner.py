@@ -291,8 +291,8 @@ def learn(
         and flags.learner_device.type == "cuda"
         and reducer.world_size == 1
         and flags.num_learner_threads == 1
-        and flags.model == "shallow"
-        and not flags.use_lstm
+        and not flags.use_lstm  # shallow AND deep capture (bench.py default);
+        # the persistent LSTM gains nothing from replay (PROFILE_r2.md)
     )
     graph = None
     static_in = None
